@@ -1,0 +1,113 @@
+"""Distributed MNIST-MLP training — the reference repo's main script,
+rebuilt on rocps (SURVEY.md §2a "main training script").
+
+Run exactly like the reference (one command per process):
+
+    python examples/dist_mnist.py --ps_hosts=127.0.0.1:2222 \
+        --worker_hosts=127.0.0.1:2223,127.0.0.1:2224 \
+        --job_name=ps --job_name=ps --task_index=0
+    python examples/dist_mnist.py ... --job_name=worker --task_index=0
+    python examples/dist_mnist.py ... --job_name=worker --task_index=1 --issync=1
+
+Sync mode (--issync): exactly-N gradient aggregation per step (rocps
+SyncEngine: bucketed reduce-scatter/all-reduce + barrier semantics of
+tf.train.SyncReplicasOptimizer). Async mode: host-DRAM PS shards.
+Data is synthetic (deterministic per global step) — no network access.
+"""
+
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+import torch  # noqa: E402
+
+import rocps  # noqa: E402
+from rocps import flags  # noqa: E402
+from rocps.models import mnist_mlp, synthetic_mnist_batch  # noqa: E402
+
+flags.DEFINE_string("ps_hosts", "", "comma-separated host:port list")
+flags.DEFINE_string("worker_hosts", "", "comma-separated host:port list")
+flags.DEFINE_string("job_name", "", "'ps' or 'worker'")
+flags.DEFINE_integer("task_index", 0, "task index within the job")
+flags.DEFINE_boolean("issync", True, "synchronous (exactly-N) aggregation")
+flags.DEFINE_integer("train_steps", 50, "number of global steps")
+flags.DEFINE_integer("batch_size", 32, "per-worker batch size")
+flags.DEFINE_float("learning_rate", 0.1, "SGD learning rate")
+flags.DEFINE_string("checkpoint_dir", "", "chief checkpoint dir (optional)")
+flags.DEFINE_integer("save_checkpoint_steps", 0, "ckpt every N steps (0=off)")
+flags.DEFINE_integer("hidden", 32, "MLP hidden width")
+flags.DEFINE_integer("seed", 42, "data/model seed")
+flags.DEFINE_string("loss_log", "", "write per-step losses (JSON) here")
+FLAGS = flags.FLAGS
+
+
+def main(_):
+    cluster = rocps.train.ClusterSpec({
+        "ps": [h for h in FLAGS.ps_hosts.split(",") if h],
+        "worker": [h for h in FLAGS.worker_hosts.split(",") if h],
+    })
+    server = rocps.train.Server(cluster, FLAGS.job_name, FLAGS.task_index)
+
+    if FLAGS.job_name == "ps":
+        if not FLAGS.issync:
+            from rocps.engine.async_engine import AsyncPSShard
+            AsyncPSShard.serve_on(server, cluster)
+        server.join()
+        return 0
+
+    num_workers = cluster.num_tasks("worker")
+    is_chief = FLAGS.task_index == 0
+    torch.manual_seed(FLAGS.seed)
+    model = mnist_mlp(hidden=FLAGS.hidden)
+    opt = rocps.train.GradientDescentOptimizer(FLAGS.learning_rate)
+    if FLAGS.issync:
+        opt = rocps.train.SyncReplicasOptimizer(
+            opt, replicas_to_aggregate=num_workers,
+            total_num_replicas=num_workers)
+    trainer = rocps.train.Trainer(
+        model, opt, cluster=cluster, job_name=FLAGS.job_name,
+        task_index=FLAGS.task_index,
+        mode="sync" if FLAGS.issync else "async")
+
+    hooks = [rocps.train.StopAtStepHook(last_step=FLAGS.train_steps),
+             rocps.train.StepCounterHook(every_n_steps=20,
+                                         batch_size=FLAGS.batch_size * num_workers)]
+    losses = []
+    global_batch = FLAGS.batch_size * num_workers
+    with rocps.train.MonitoredTrainingSession(
+            trainer, is_chief=is_chief, hooks=hooks,
+            checkpoint_dir=FLAGS.checkpoint_dir or None,
+            save_checkpoint_steps=(FLAGS.save_checkpoint_steps or None),
+            save_checkpoint_secs=(None if FLAGS.save_checkpoint_steps else 600),
+    ) as sess:
+        while not sess.should_stop():
+            # deterministic synthetic batch for this global step; each
+            # worker takes its slice (the reference feeds next_batch())
+            g = torch.Generator().manual_seed(FLAGS.seed + trainer.global_step)
+            x, y = synthetic_mnist_batch(global_batch, generator=g)
+            lo = FLAGS.task_index * FLAGS.batch_size
+            xb, yb = x[lo:lo + FLAGS.batch_size], y[lo:lo + FLAGS.batch_size]
+            loss = sess.run(lambda: trainer.model.loss(xb, yb))
+            losses.append(float(loss))
+
+    if FLAGS.loss_log:
+        with open(FLAGS.loss_log, "w") as f:
+            json.dump(losses, f)
+    # every worker reports done; the chief shuts the cluster down only
+    # once all workers finished (PS must keep serving until then).
+    import time as _time
+    trainer.control.incr("workers_done")
+    if is_chief:
+        deadline = _time.time() + 300
+        while (trainer.control.read_counter("workers_done") < num_workers
+               and _time.time() < deadline):
+            _time.sleep(0.1)
+        from rocps.cluster import shutdown_cluster
+        shutdown_cluster(cluster)
+    return 0
+
+
+if __name__ == "__main__":
+    rocps.app.run(main)

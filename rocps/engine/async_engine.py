@@ -1,0 +1,373 @@
+"""Async-SGD parameter-server engine — capability C5 of SURVEY.md §2b.
+
+Reference semantics (§3.2): each worker independently pushes gradients;
+the optimizer kernels run *on the PS device* immediately (no
+coordination, ``use_locking=False`` — parameter reads may be torn /
+stale by design); the worker then pulls fresh parameters. A worker's
+train op completes only when its own apply finished (the gRPC RunGraph
+round trip), but workers never wait for each other.
+
+MI355X-native recast (single node, 8 GPU workers + host-DRAM PS):
+- Each PS task owns a *shard* of the flat parameter space in a
+  POSIX shared-memory segment (/dev/shm), with per-worker gradient
+  slots and seq/ack words.
+- Workers mmap the segment: gradient push is one device→host copy per
+  shard (hipMemcpyAsync once the region is hipHostRegister'ed; plain
+  copy on CPU), then a seq bump; parameter pull is one host→device
+  copy. No per-tensor RPC, no serialization — the copy engine moves
+  the bytes over PCIe/IF while compute streams stay busy.
+- The PS apply thread spins on seq!=ack, applies the (fused CPU)
+  optimizer per worker push, bumps version + global_step. Staleness
+  accounting: each push carries the worker's local step; the shard
+  drops pushes older than ``version - staleness_bound`` when a bound
+  is configured (the SyncReplicasOptimizer stale-drop rule, C4/C5).
+
+Wire bootstrap (control plane): the chief publishes a manifest and
+init params; each PS task serves ``init_shard`` and publishes its shm
+path; workers attach.
+"""
+
+from __future__ import annotations
+
+import json
+import mmap
+import os
+import struct
+import threading
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from rocps.cluster import ClusterSpec, ControlClient, Server
+from rocps.placement import plan_shards
+from rocps.train.coordinator import Coordinator
+
+__all__ = ["AsyncPSShard", "AsyncWorkerEngine"]
+
+_MAGIC = 0x52C0_5A51
+_ALIGN = 256
+
+
+def _np_dtype(t: torch.dtype):
+    return {torch.float32: np.float32, torch.float64: np.float64,
+            torch.bfloat16: np.uint16, torch.float16: np.float16}[t]
+
+
+class _ShmLayout:
+    """header int64[8 + 2*W] | params | grads[W]  (offsets 256-aligned).
+
+    header: [magic, version(applies), num_workers, numel, dtype_code,
+             global_step, reserved, reserved, then per worker: seq, ack]
+    """
+
+    def __init__(self, numel: int, num_workers: int, itemsize: int):
+        self.numel = numel
+        self.num_workers = num_workers
+        self.header_words = 8 + 2 * num_workers
+        self.params_off = ((self.header_words * 8 + _ALIGN - 1)
+                           // _ALIGN * _ALIGN)
+        grad_bytes = numel * itemsize
+        self.grad_off0 = ((self.params_off + grad_bytes + _ALIGN - 1)
+                          // _ALIGN * _ALIGN)
+        self.grad_stride = (grad_bytes + _ALIGN - 1) // _ALIGN * _ALIGN
+        self.total = self.grad_off0 + self.grad_stride * num_workers
+
+    def grad_off(self, w: int) -> int:
+        return self.grad_off0 + w * self.grad_stride
+
+
+class _ShmSeg:
+    """A mapped shared-memory segment with torch views."""
+
+    def __init__(self, path: str, layout: _ShmLayout, dtype: torch.dtype,
+                 create: bool):
+        self.path = path
+        self.layout = layout
+        self.dtype = dtype
+        flags = os.O_RDWR | (os.O_CREAT if create else 0)
+        self.fd = os.open(path, flags, 0o600)
+        if create:
+            os.ftruncate(self.fd, layout.total)
+        self.mm = mmap.mmap(self.fd, layout.total)
+        self.buf = np.frombuffer(self.mm, dtype=np.uint8)
+        self.header = torch.from_numpy(
+            np.frombuffer(self.mm, dtype=np.int64,
+                          count=layout.header_words, offset=0))
+        npdt = _np_dtype(dtype)
+        params_np = np.frombuffer(self.mm, dtype=npdt, count=layout.numel,
+                                  offset=layout.params_off)
+        self.params = torch.from_numpy(params_np)
+        if dtype == torch.bfloat16:
+            self.params = self.params.view(torch.bfloat16)
+        self.grads: List[torch.Tensor] = []
+        for w in range(layout.num_workers):
+            g_np = np.frombuffer(self.mm, dtype=npdt, count=layout.numel,
+                                 offset=layout.grad_off(w))
+            g = torch.from_numpy(g_np)
+            if dtype == torch.bfloat16:
+                g = g.view(torch.bfloat16)
+            self.grads.append(g)
+
+    def close(self) -> None:
+        del self.header, self.params, self.grads, self.buf
+        self.mm.close()
+        os.close(self.fd)
+
+
+# ---------------------------------------------------------------------------
+# PS side
+# ---------------------------------------------------------------------------
+
+class AsyncPSShard:
+    """Runs inside a ps task process: hosts one shard + its apply thread."""
+
+    def __init__(self, server: Server, cluster: ClusterSpec,
+                 shard_index: int):
+        self.server = server
+        self.cluster = cluster
+        self.shard_index = shard_index
+        self.master = ControlClient(cluster.master_address)
+        self.num_workers = cluster.num_tasks("worker")
+        self.coord = Coordinator()
+        self.seg: Optional[_ShmSeg] = None
+        self.optimizer = None
+        self._apply_thread: Optional[threading.Thread] = None
+        server.register_handler("init_shard", self._on_init_shard)
+        server.register_handler("shard_stats", self._on_stats)
+        self.applied = 0
+        self.dropped_stale = 0
+        self.staleness_bound: Optional[int] = None
+
+    @staticmethod
+    def serve_on(server: Server, cluster: ClusterSpec) -> "AsyncPSShard":
+        idx = cluster.task_indices("ps").index(server.task_index)
+        return AsyncPSShard(server, cluster, idx)
+
+    # -- op handlers ----------------------------------------------------
+    def _on_init_shard(self, head: Dict, payload: bytes):
+        numel = int(head["numel"])
+        dtype = getattr(torch, head["dtype"])
+        itemsize = torch.empty(0, dtype=dtype).element_size()
+        layout = _ShmLayout(numel, self.num_workers, itemsize)
+        path = f"/dev/shm/rocps_shard{self.shard_index}_{os.getpid()}"
+        if self.seg is not None:
+            self.seg.close()
+        self.seg = _ShmSeg(path, layout, dtype, create=True)
+        h = self.seg.header
+        h.zero_()
+        h[0] = _MAGIC
+        h[2] = self.num_workers
+        h[3] = numel
+        # initial params from chief
+        src = np.frombuffer(payload, dtype=_np_dtype(dtype), count=numel)
+        t = torch.from_numpy(src.copy())
+        if dtype == torch.bfloat16:
+            t = t.view(torch.bfloat16)
+        self.seg.params.copy_(t)
+        # optimizer from manifest
+        opt_spec = head.get("optimizer", {"type": "sgd", "lr": 0.01})
+        self.optimizer = _build_optimizer(opt_spec)
+        self.staleness_bound = head.get("staleness_bound")
+        if self._apply_thread is None:
+            self._apply_thread = threading.Thread(
+                target=self._apply_loop, daemon=True,
+                name=f"rocps-ps-apply-{self.shard_index}")
+            self._apply_thread.start()
+        self.master.put(f"async_shm_{self.shard_index}",
+                        {"path": path, "numel": numel,
+                         "dtype": head["dtype"]})
+        return {"ok": True, "path": path}, b""
+
+    def _on_stats(self, head, payload):
+        return {"ok": True, "applied": self.applied,
+                "dropped_stale": self.dropped_stale,
+                "version": int(self.seg.header[1]) if self.seg else 0}, b""
+
+    # -- the apply loop --------------------------------------------------
+    def _apply_loop(self) -> None:
+        seg = self.seg
+        h = seg.header
+        W = self.num_workers
+        with self.coord.stop_on_exception():
+            while not (self.coord.should_stop()
+                       or self.server._shutdown_evt.is_set()):
+                progressed = False
+                for w in range(W):
+                    seq, ack = int(h[8 + 2 * w]), int(h[8 + 2 * w + 1])
+                    if seq != ack:
+                        version = int(h[1])
+                        # seq encodes (worker local step); stale-drop rule
+                        if (self.staleness_bound is not None
+                                and version - seq > self.staleness_bound):
+                            self.dropped_stale += 1
+                        else:
+                            self.optimizer.apply_flat(
+                                seg.params, seg.grads[w], slot="shard")
+                            self.applied += 1
+                            h[1] = version + 1
+                        h[5] += 1  # global_step mirror
+                        h[8 + 2 * w + 1] = seq  # ack
+                        progressed = True
+                if not progressed:
+                    time.sleep(0.0002)
+
+
+def _build_optimizer(spec: Dict):
+    from rocps.train.optimizer import (AdamOptimizer,
+                                       GradientDescentOptimizer,
+                                       MomentumOptimizer)
+    t = spec.get("type", "sgd")
+    if t == "sgd":
+        return GradientDescentOptimizer(spec.get("lr", 0.01))
+    if t == "momentum":
+        return MomentumOptimizer(spec.get("lr", 0.01),
+                                 spec.get("momentum", 0.9),
+                                 spec.get("nesterov", False))
+    if t == "adam":
+        o = AdamOptimizer(spec.get("lr", 1e-3), spec.get("beta1", 0.9),
+                          spec.get("beta2", 0.999), spec.get("eps", 1e-8))
+        return o
+    raise ValueError(f"unknown optimizer type {t!r}")
+
+
+def optimizer_spec(opt) -> Dict:
+    from rocps.train.optimizer import (AdamOptimizer,
+                                       GradientDescentOptimizer,
+                                       MomentumOptimizer)
+    base = getattr(opt, "base", opt)
+    if isinstance(base, MomentumOptimizer):
+        return {"type": "momentum", "lr": base.lr, "momentum": base.momentum,
+                "nesterov": base.use_nesterov}
+    if isinstance(base, AdamOptimizer):
+        return {"type": "adam", "lr": base.lr, "beta1": base.beta1,
+                "beta2": base.beta2, "eps": base.epsilon}
+    if isinstance(base, GradientDescentOptimizer):
+        return {"type": "sgd", "lr": base.lr}
+    return {"type": "sgd", "lr": getattr(base, "lr", 0.01)}
+
+
+# ---------------------------------------------------------------------------
+# Worker side
+# ---------------------------------------------------------------------------
+
+class AsyncWorkerEngine:
+    """Worker-side engine: flat per-shard grad buffers; push/ack/pull."""
+
+    def __init__(self, model: torch.nn.Module, cluster: ClusterSpec,
+                 control: ControlClient, task_index: int,
+                 device: torch.device, num_shards: Optional[int] = None,
+                 strategy: str = "greedy"):
+        self.model = model
+        self.cluster = cluster
+        self.control = control
+        self.task_index = task_index
+        self.device = device
+        self.num_shards = num_shards or cluster.num_tasks("ps")
+        named = [(n, p) for n, p in model.named_parameters()
+                 if p.requires_grad]
+        self.plan = plan_shards(named, self.num_shards, strategy=strategy)
+        # group params by shard, keep deterministic order
+        self.shard_params: List[List] = [[] for _ in range(self.num_shards)]
+        for n, p in named:
+            self.shard_params[self.plan.shard_of(n)].append((n, p))
+        self.dtype = named[0][1].dtype
+        # local flat param/grad buffers per shard; param/grad views wired
+        self.flat_params: List[torch.Tensor] = []
+        self.flat_grads: List[torch.Tensor] = []
+        for s in range(self.num_shards):
+            numel = sum(p.numel() for _, p in self.shard_params[s])
+            fp = torch.zeros(numel, dtype=self.dtype, device=device)
+            fg = torch.zeros(numel, dtype=self.dtype, device=device)
+            off = 0
+            for n, p in self.shard_params[s]:
+                k = p.numel()
+                fp[off:off + k] = p.data.reshape(-1)
+                p.data = fp[off:off + k].view_as(p.data)
+                p.grad = fg[off:off + k].view_as(p.data)
+                off += k
+            self.flat_params.append(fp)
+            self.flat_grads.append(fg)
+        self.segs: List[Optional[_ShmSeg]] = [None] * self.num_shards
+        self.local_step = 0
+        self.is_chief = task_index == 0
+        self._attached = False
+        self.last_global_step = 0
+
+    # -- bootstrap -------------------------------------------------------
+    def initialize(self, optimizer) -> None:
+        if self.is_chief:
+            for s in range(self.num_shards):
+                ps_addr = self.cluster.task_address(
+                    "ps", self.cluster.task_indices("ps")[s])
+                c = ControlClient(ps_addr)
+                flat = self.flat_params[s].detach().cpu()
+                view = (flat.view(torch.uint16) if self.dtype == torch.bfloat16
+                        else flat)
+                payload = view.numpy().tobytes()
+                r = c.call("init_shard", payload=payload,
+                           numel=flat.numel(),
+                           dtype=str(self.dtype).replace("torch.", ""),
+                           optimizer=optimizer_spec(optimizer))
+                if not r.get("ok"):
+                    raise RuntimeError(f"init_shard failed: {r}")
+                c.close()
+        self._attach()
+
+    def _attach(self) -> None:
+        for s in range(self.num_shards):
+            info = self.control.get(f"async_shm_{s}", blocking=True,
+                                    timeout=120.0)
+            dtype = getattr(torch, info["dtype"])
+            itemsize = torch.empty(0, dtype=dtype).element_size()
+            layout = _ShmLayout(int(info["numel"]),
+                                self.cluster.num_tasks("worker"), itemsize)
+            self.segs[s] = _ShmSeg(info["path"], layout, dtype, create=False)
+        self._attached = True
+        self.pull()
+
+    # -- engine interface (used by Trainer) ------------------------------
+    def broadcast_parameters(self, src: int = 0) -> None:
+        pass  # params come from the PS pull
+
+    def zero_grad(self) -> None:
+        for g in self.flat_grads:
+            g.zero_()
+
+    def step(self, loss: torch.Tensor, optimizer) -> None:
+        if not self._attached:
+            self.initialize(optimizer)
+        loss.backward()
+        self.push_and_pull()
+
+    def push_and_pull(self) -> None:
+        self.local_step += 1
+        w = self.task_index
+        # push grads to every shard, bump seq
+        for s, seg in enumerate(self.segs):
+            seg.grads[w].copy_(self.flat_grads[s].cpu()
+                               if self.device.type != "cpu"
+                               else self.flat_grads[s])
+            seg.header[8 + 2 * w] = self.local_step
+        # wait for own applies (the reference's RunGraph round trip)
+        deadline = time.time() + 120.0
+        for seg in self.segs:
+            while int(seg.header[8 + 2 * w + 1]) != self.local_step:
+                if time.time() > deadline:
+                    raise TimeoutError("PS apply timeout (shard dead?)")
+                time.sleep(0.0002)
+        # pull fresh (possibly concurrently-updated) params
+        self.pull()
+        self.last_global_step = int(self.segs[0].header[5])
+
+    def pull(self) -> None:
+        for s, seg in enumerate(self.segs):
+            self.flat_params[s].copy_(seg.params.to(self.device)
+                                      if self.device.type != "cpu"
+                                      else seg.params)
+
+    def detach(self) -> None:
+        for seg in self.segs:
+            if seg is not None:
+                seg.close()

@@ -1,0 +1,104 @@
+"""MonitoredTrainingSession + hooks + Saver tests (C12/C13)."""
+
+import os
+
+import pytest
+import torch
+
+from rocps.models import mnist_mlp, synthetic_mnist_batch
+from rocps.train import (AdamOptimizer, GradientDescentOptimizer,
+                         MomentumOptimizer, MonitoredTrainingSession,
+                         NanTensorHook, Saver, StepCounterHook, StopAtStepHook,
+                         Trainer, latest_checkpoint)
+
+
+def _run(trainer, steps, ckdir=None, seed=5, save_every=None, hooks=()):
+    g = torch.Generator().manual_seed(seed)
+    losses = []
+    with MonitoredTrainingSession(
+            trainer, checkpoint_dir=ckdir,
+            save_checkpoint_steps=save_every,
+            save_checkpoint_secs=None if save_every else 600,
+            hooks=[StopAtStepHook(last_step=steps), *hooks]) as sess:
+        while not sess.should_stop():
+            gg = torch.Generator().manual_seed(seed + trainer.global_step)
+            x, y = synthetic_mnist_batch(16, generator=gg)
+            losses.append(float(sess.run(lambda: trainer.model.loss(x, y))))
+    return losses
+
+
+def test_stop_at_step():
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    losses = _run(tr, 7)
+    assert tr.global_step == 7 and len(losses) == 7
+
+
+def test_nan_hook_raises():
+    class BadModel(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.w = torch.nn.Parameter(torch.ones(1))
+
+        def loss(self, x, y):
+            return self.w.sum() * float("nan")
+
+    tr = Trainer(BadModel(), GradientDescentOptimizer(0.1))
+    with pytest.raises(RuntimeError, match="NaN"):
+        _run(tr, 3, hooks=[NanTensorHook()])
+
+
+def test_checkpoint_resume_exact_continuation(tmp_path):
+    """Train 30 straight vs train 15 + resume + 15: identical params."""
+    ckdir = str(tmp_path / "ck")
+    torch.manual_seed(11)
+    mA = mnist_mlp(hidden=16)
+    trA = Trainer(mA, MomentumOptimizer(0.1, 0.9))
+    _run(trA, 30)
+
+    torch.manual_seed(11)
+    mB = mnist_mlp(hidden=16)
+    trB = Trainer(mB, MomentumOptimizer(0.1, 0.9))
+    _run(trB, 15, ckdir=ckdir, save_every=15)
+    assert latest_checkpoint(ckdir).endswith("-15")
+
+    torch.manual_seed(99)  # different init — restore must overwrite it
+    mC = mnist_mlp(hidden=16)
+    trC = Trainer(mC, MomentumOptimizer(0.1, 0.9))
+    _run(trC, 30, ckdir=ckdir, save_every=15)  # restores at 15, runs to 30
+    assert trC.global_step == 30
+    pA = dict(mA.named_parameters())
+    pC = dict(mC.named_parameters())
+    for k in pA:
+        assert torch.allclose(pA[k], pC[k], atol=1e-7), k
+
+
+def test_checkpoint_includes_optimizer_state(tmp_path):
+    ckdir = str(tmp_path / "ck2")
+    tr = Trainer(mnist_mlp(hidden=8), AdamOptimizer(1e-3))
+    _run(tr, 5, ckdir=ckdir, save_every=5)
+    from rocps.checkpoint.bundle import list_bundle
+    names = {e["name"] for e in list_bundle(latest_checkpoint(ckdir))}
+    assert "global_step" in names
+    assert any(n.startswith("optimizer/") and n.endswith("/m") for n in names)
+    assert any(n.startswith("optimizer/") and n.endswith("/v") for n in names)
+
+
+def test_max_to_keep_gc(tmp_path):
+    ckdir = str(tmp_path / "ck3")
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    saver = Saver(tr, max_to_keep=2)
+    for s in [1, 2, 3]:
+        tr.global_step = s
+        saver.save(None, os.path.join(ckdir, "model.ckpt"), global_step=s)
+    files = os.listdir(ckdir)
+    assert "model.ckpt-1.index" not in files
+    assert "model.ckpt-2.index" in files and "model.ckpt-3.index" in files
+    assert latest_checkpoint(ckdir).endswith("model.ckpt-3")
+    assert Saver.last_checkpoints(ckdir) == ["model.ckpt-2", "model.ckpt-3"]
+
+
+def test_step_counter_hook():
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    sc = StepCounterHook(every_n_steps=2, batch_size=16)
+    _run(tr, 6, hooks=[sc])
+    assert sc.last_steps_per_sec is not None and sc.last_steps_per_sec > 0
