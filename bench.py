@@ -1,0 +1,181 @@
+"""rocps flagship benchmark — BASELINE.json's headline metric:
+images/sec (whole node), ResNet-50 sync-SGD, bf16, synthetic data,
+random-init weights, at 1..8 MI355X.
+
+Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi-GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Each rank = one GPU worker with its co-located PS shard (reduce-scatter
++ fused-optimizer + all-gather over RCCL/xGMI). Rank 0 prints ONE JSON
+line with the whole-job aggregate.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=256,
+                    help="per-GPU batch size")
+    ap.add_argument("--model", type=str, default="resnet50",
+                    choices=["resnet50", "wide_resnet101_2", "bert_base",
+                             "mnist_mlp"])
+    ap.add_argument("--mode", type=str, default="sync",
+                    choices=["sync", "async"])
+    ap.add_argument("--bucket-mb", type=float, default=25.0)
+    ap.add_argument("--seq-len", type=int, default=128)
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["bf16", "fp32"])
+    ap.add_argument("--device", type=str, default=None)
+    return ap.parse_args()
+
+
+def build(args, device, dtype):
+    """Returns (model, optimizer, make_batch) for the chosen config."""
+    from rocps.train.optimizer import AdamOptimizer, MomentumOptimizer
+    torch.manual_seed(1234)
+    if args.model in ("resnet50", "wide_resnet101_2"):
+        from rocps.models.resnet import resnet50, wide_resnet101_2
+        m = resnet50() if args.model == "resnet50" else wide_resnet101_2()
+        m = m.to(device)
+        if dtype == torch.bfloat16:
+            m = m.bfloat16()
+        m = m.to(memory_format=torch.channels_last)
+        opt = MomentumOptimizer(0.256, 0.875)
+
+        def make_batch(i):
+            g = torch.Generator(device="cpu").manual_seed(i)
+            x = torch.randn(args.batch, 3, 224, 224, generator=g).to(dtype)
+            y = torch.randint(0, 1000, (args.batch,), generator=g)
+            return (x.to(device, non_blocking=True)
+                    .to(memory_format=torch.channels_last),
+                    y.to(device, non_blocking=True))
+
+        return m, opt, make_batch
+    if args.model == "bert_base":
+        from rocps.models.bert import bert_base
+        m = bert_base().to(device)
+        if dtype == torch.bfloat16:
+            m = m.bfloat16()
+        opt = AdamOptimizer(1e-4)
+
+        def make_batch(i):
+            g = torch.Generator(device="cpu").manual_seed(i)
+            ids = torch.randint(0, 30522, (args.batch, args.seq_len),
+                                generator=g)
+            labels = torch.randint(0, 30522, (args.batch, args.seq_len),
+                                   generator=g)
+            return ids.to(device), labels.to(device)
+
+        return m, opt, make_batch
+    # mnist_mlp (CPU-friendly sanity config)
+    from rocps.models import mnist_mlp
+
+    m = mnist_mlp(hidden=128).to(device)
+    opt = MomentumOptimizer(0.05, 0.9)
+
+    def make_batch(i):
+        g = torch.Generator(device="cpu").manual_seed(i)
+        x = torch.rand(args.batch, 784, generator=g)
+        y = torch.randint(0, 10, (args.batch,), generator=g)
+        return x.to(device), y.to(device)
+
+    return m, opt, make_batch
+
+
+def main():
+    args = parse_args()
+    from rocps.engine.comm import Comm
+    from rocps.engine.sync_engine import SyncEngine
+
+    Comm.tune_env()
+    comm = Comm.from_env()
+    device = (torch.device(args.device) if args.device else comm.device)
+    use_cuda = device.type == "cuda"
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    if not use_cuda and args.dtype == "bf16" and args.model != "mnist_mlp":
+        dtype = torch.float32  # CPU sanity runs in fp32
+
+    model, opt, make_batch = build(args, device, dtype)
+    engine = SyncEngine(model, comm, bucket_cap_mb=args.bucket_mb)
+    engine.broadcast_parameters(src=0)
+
+    n_batches = 4
+    batches = [make_batch(1000 * comm.rank + i) for i in range(n_batches)]
+
+    def step(i):
+        x, y = batches[i % n_batches]
+        engine.zero_grad()
+        if hasattr(opt, "begin_step"):
+            opt.begin_step()
+        loss = model.loss(x, y)
+        engine.step(loss, opt)
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (conservative whole-job time)
+    t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+    if comm.is_distributed:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    n = comm.world_size
+    global_batch = args.batch * n
+    if args.model == "bert_base":
+        value = global_batch * args.seq_len * args.steps / elapsed
+        metric, unit = "tokens/sec", "tokens/s"
+    else:
+        value = global_batch * args.steps / elapsed
+        metric, unit = "images/sec", "images/s"
+    if comm.rank == 0:
+        out = {
+            "metric": metric,
+            "value": round(value, 1),
+            "unit": unit,
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference published no numbers
+            "dtype": ("bf16" if next(model.parameters()).dtype
+                      == torch.bfloat16 else "fp32"),
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": args.seq_len if args.model == "bert_base" else None,
+                "parallelism": f"dp{n}",
+                "mode": args.mode,
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if comm.is_distributed:
+        comm.destroy()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,228 @@
+// Fused BatchNorm(+ReLU) forward/backward on NHWC — C9's BN slice.
+//
+// Reference behavior: TF FusedBatchNormV3[Grad] + Relu[Grad]
+// (tensorflow/core/kernels/fused_batch_norm_op.cc, relu_op.cc) on every
+// conv output of ResNet/W-ResNet. Recast for CDNA4 NHWC:
+//
+//   fwd train: stats kernel (per-channel sum/sumsq, fp32, per-thread
+//     register accumulation over a row range + 2 device atomicAdds per
+//     thread) → per-channel scale/shift folded on the small C-sized
+//     tensors → apply kernel y = relu(x*scale[c]+shift[c]) vectorized
+//     8 channels/lane (bf16 ushort8 = 16 B, Guideline 13).
+//   bwd: reduce kernel (sum_dy, sum_dy*xhat with the ReLU mask folded)
+//     → per-channel constants → apply kernel dx = A[c]*dy_m + B[c]*x +
+//     D[c]. dgamma/dbeta fall out of the reduction.
+//
+// Layout contract: x is channels_last → memory is (M=N*H*W) rows × C
+// contiguous channels. Lanes map to adjacent channels (coalesced); the
+// per-channel constant arrays are C-sized and L2-resident.
+
+#include "common.h"
+#include "launchers.h"
+
+namespace {
+
+// ---- per-channel reduction over rows (stats + bwd reduce) ----------------
+template <typename T>
+__global__ void bn_stats_kernel(const T* __restrict__ x,
+                                float* __restrict__ gsum,
+                                float* __restrict__ gsumsq, long M, int C) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, ss = 0.f;
+  for (long r = blockIdx.x; r < M; r += gridDim.x) {
+    float v = to_f32<T>(x[r * C + c]);
+    s += v;
+    ss = fmaf(v, v, ss);
+  }
+  atomicAdd(&gsum[c], s);
+  atomicAdd(&gsumsq[c], ss);
+}
+
+template <typename T, bool RELU>
+__global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ dy,
+                                     const T* __restrict__ y,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     float* __restrict__ gsum_dy,
+                                     float* __restrict__ gsum_dyxh,
+                                     long M, int C) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mu = mean[c], is = invstd[c];
+  float s1 = 0.f, s2 = 0.f;
+  for (long r = blockIdx.x; r < M; r += gridDim.x) {
+    long i = r * C + c;
+    float d = to_f32<T>(dy[i]);
+    if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
+    float xh = (to_f32<T>(x[i]) - mu) * is;
+    s1 += d;
+    s2 = fmaf(d, xh, s2);
+  }
+  atomicAdd(&gsum_dy[c], s1);
+  atomicAdd(&gsum_dyxh[c], s2);
+}
+
+// ---- elementwise applies --------------------------------------------------
+struct alignas(16) u16x8 { unsigned short v[8]; };
+
+template <typename T, bool RELU, bool VEC8>
+__global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                const float* __restrict__ scale,
+                                const float* __restrict__ shift,
+                                long total, int C) {
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  if (VEC8) {
+    long nvec = total / 8;
+    for (long i = tid; i < nvec; i += stride) {
+      long base = i * 8;
+      int c0 = (int)(base % C);
+      float out[8];
+#pragma unroll
+      for (int k = 0; k < 8; k++) {
+        float v = to_f32<T>(x[base + k]);
+        int c = c0 + k;  // C % 8 == 0 → the 8 lanes stay inside one row
+        v = fmaf(v, scale[c], shift[c]);
+        out[k] = RELU ? fmaxf(v, 0.f) : v;
+      }
+#pragma unroll
+      for (int k = 0; k < 8; k++) y[base + k] = from_f32<T>(out[k]);
+    }
+  } else {
+    for (long i = tid; i < total; i += stride) {
+      int c = (int)(i % C);
+      float v = fmaf(to_f32<T>(x[i]), scale[c], shift[c]);
+      y[i] = from_f32<T>(RELU ? fmaxf(v, 0.f) : v);
+    }
+  }
+}
+
+template <typename T, bool RELU>
+__global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ dy,
+                                    const T* __restrict__ y,
+                                    T* __restrict__ dx,
+                                    const float* __restrict__ A,
+                                    const float* __restrict__ B,
+                                    const float* __restrict__ D,
+                                    long total, int C) {
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = tid; i < total; i += stride) {
+    int c = (int)(i % C);
+    float d = to_f32<T>(dy[i]);
+    if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
+    float r = fmaf(A[c], d, fmaf(B[c], to_f32<T>(x[i]), D[c]));
+    dx[i] = from_f32<T>(r);
+  }
+}
+
+}  // namespace
+
+// ---- launchers ------------------------------------------------------------
+template <typename T>
+static void _stats(const T* x, float* gsum, float* gsumsq, long M, int C,
+                   hipStream_t stream) {
+  int block = 256;
+  dim3 grid((unsigned)((M + 255) / 256 < 64 ? ((M + 255) / 256 ?: 1) : 64),
+            (C + block - 1) / block);
+  hipLaunchKernelGGL((bn_stats_kernel<T>), grid, dim3(block), 0, stream, x,
+                     gsum, gsumsq, M, C);
+}
+void launch_bn_stats_f32(const float* x, float* s, float* ss, long M, int C,
+                         hipStream_t st) { _stats(x, s, ss, M, C, st); }
+void launch_bn_stats_bf16(const rbf16* x, float* s, float* ss, long M, int C,
+                          hipStream_t st) {
+  _stats(reinterpret_cast<const bf16*>(x), s, ss, M, C, st);
+}
+
+template <typename T>
+static void _apply(const T* x, T* y, const float* scale, const float* shift,
+                   long total, int C, bool relu, hipStream_t stream) {
+  int block = 256;
+  int grid = rocps_grid_for((total + 7) / 8, block);
+  bool vec8 = (C % 8 == 0);
+  if (relu && vec8)
+    hipLaunchKernelGGL((bn_apply_kernel<T, true, true>), dim3(grid),
+                       dim3(block), 0, stream, x, y, scale, shift, total, C);
+  else if (relu)
+    hipLaunchKernelGGL((bn_apply_kernel<T, true, false>), dim3(grid),
+                       dim3(block), 0, stream, x, y, scale, shift, total, C);
+  else if (vec8)
+    hipLaunchKernelGGL((bn_apply_kernel<T, false, true>), dim3(grid),
+                       dim3(block), 0, stream, x, y, scale, shift, total, C);
+  else
+    hipLaunchKernelGGL((bn_apply_kernel<T, false, false>), dim3(grid),
+                       dim3(block), 0, stream, x, y, scale, shift, total, C);
+}
+void launch_bn_apply_f32(const float* x, float* y, const float* sc,
+                         const float* sh, long total, int C, bool relu,
+                         hipStream_t st) { _apply(x, y, sc, sh, total, C, relu, st); }
+void launch_bn_apply_bf16(const rbf16* x, rbf16* y, const float* sc,
+                          const float* sh, long total, int C, bool relu,
+                          hipStream_t st) {
+  _apply(reinterpret_cast<const bf16*>(x), reinterpret_cast<bf16*>(y), sc, sh,
+         total, C, relu, st);
+}
+
+template <typename T>
+static void _bwd_reduce(const T* x, const T* dy, const T* y,
+                        const float* mean, const float* invstd, float* s1,
+                        float* s2, long M, int C, bool relu,
+                        hipStream_t stream) {
+  int block = 256;
+  dim3 grid((unsigned)((M + 255) / 256 < 64 ? ((M + 255) / 256 ?: 1) : 64),
+            (C + block - 1) / block);
+  if (relu)
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid, dim3(block), 0,
+                       stream, x, dy, y, mean, invstd, s1, s2, M, C);
+  else
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid, dim3(block), 0,
+                       stream, x, dy, y, mean, invstd, s1, s2, M, C);
+}
+void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
+                              const float* mean, const float* invstd,
+                              float* s1, float* s2, long M, int C, bool relu,
+                              hipStream_t st) {
+  _bwd_reduce(x, dy, y, mean, invstd, s1, s2, M, C, relu, st);
+}
+void launch_bn_bwd_reduce_bf16(const rbf16* x, const rbf16* dy,
+                               const rbf16* y, const float* mean,
+                               const float* invstd, float* s1, float* s2,
+                               long M, int C, bool relu, hipStream_t st) {
+  _bwd_reduce(reinterpret_cast<const bf16*>(x),
+              reinterpret_cast<const bf16*>(dy),
+              reinterpret_cast<const bf16*>(y), mean, invstd, s1, s2, M, C,
+              relu, st);
+}
+
+template <typename T>
+static void _bwd_apply(const T* x, const T* dy, const T* y, T* dx,
+                       const float* A, const float* B, const float* D,
+                       long total, int C, bool relu, hipStream_t stream) {
+  int block = 256;
+  int grid = rocps_grid_for(total, block);
+  if (relu)
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, true>), dim3(grid),
+                       dim3(block), 0, stream, x, dy, y, dx, A, B, D, total, C);
+  else
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, false>), dim3(grid),
+                       dim3(block), 0, stream, x, dy, y, dx, A, B, D, total, C);
+}
+void launch_bn_bwd_apply_f32(const float* x, const float* dy, const float* y,
+                             float* dx, const float* A, const float* B,
+                             const float* D, long total, int C, bool relu,
+                             hipStream_t st) {
+  _bwd_apply(x, dy, y, dx, A, B, D, total, C, relu, st);
+}
+void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy,
+                              const rbf16* y, rbf16* dx, const float* A,
+                              const float* B, const float* D, long total,
+                              int C, bool relu, hipStream_t st) {
+  _bwd_apply(reinterpret_cast<const bf16*>(x),
+             reinterpret_cast<const bf16*>(dy),
+             reinterpret_cast<const bf16*>(y), reinterpret_cast<bf16*>(dx), A,
+             B, D, total, C, relu, st);
+}

@@ -1,0 +1,264 @@
+// rocps._hipops — torch bindings for the hand-written CDNA4 kernels.
+//
+// Every function validates device/dtype/layout and launches on the
+// current HIP stream; per-channel BN folds run as C-sized torch ops on
+// device (L2-resident, negligible next to the M×C passes).
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+
+#include "launchers.h"
+
+namespace {
+
+hipStream_t cur_stream(const at::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+void check_flat(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+rbf16* bf16_ptr(const at::Tensor& t) {
+  return reinterpret_cast<rbf16*>(t.data_ptr<at::BFloat16>());
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// fused optimizers (C8)
+// ---------------------------------------------------------------------------
+
+void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor master, double lr,
+               bool has_master) {
+  check_flat(p, "params");
+  check_flat(g, "grads");
+  TORCH_CHECK(p.numel() == g.numel());
+  c10::hip::OptionalHIPGuard guard(p.device());
+  long n = p.numel();
+  auto st = cur_stream(p);
+  if (p.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(!has_master || master.scalar_type() == at::kFloat);
+    launch_sgd_bf16(bf16_ptr(p), bf16_ptr(g),
+                    has_master ? master.data_ptr<float>() : nullptr,
+                    (float)lr, n, has_master, st);
+  } else {
+    TORCH_CHECK(p.scalar_type() == at::kFloat, "sgd: fp32/bf16 only");
+    launch_sgd_f32(p.data_ptr<float>(), g.data_ptr<float>(),
+                   p.data_ptr<float>(), (float)lr, n, false, st);
+  }
+}
+
+void fused_momentum(at::Tensor p, at::Tensor g, at::Tensor master,
+                    at::Tensor mom, double lr, double mu, bool nesterov,
+                    bool has_master) {
+  check_flat(p, "params");
+  check_flat(g, "grads");
+  check_flat(mom, "momentum");
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == mom.numel());
+  TORCH_CHECK(mom.scalar_type() == at::kFloat);
+  c10::hip::OptionalHIPGuard guard(p.device());
+  long n = p.numel();
+  auto st = cur_stream(p);
+  if (p.scalar_type() == at::kBFloat16) {
+    launch_momentum_bf16(bf16_ptr(p), bf16_ptr(g),
+                         has_master ? master.data_ptr<float>() : nullptr,
+                         mom.data_ptr<float>(), (float)lr, (float)mu,
+                         nesterov, n, has_master, st);
+  } else {
+    launch_momentum_f32(p.data_ptr<float>(), g.data_ptr<float>(),
+                        p.data_ptr<float>(), mom.data_ptr<float>(), (float)lr,
+                        (float)mu, nesterov, n, false, st);
+  }
+}
+
+void fused_adam(at::Tensor p, at::Tensor g, at::Tensor master, at::Tensor m,
+                at::Tensor v, double alpha, double b1, double b2, double eps,
+                bool has_master) {
+  check_flat(p, "params");
+  check_flat(g, "grads");
+  TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel()
+              && p.numel() == v.numel());
+  c10::hip::OptionalHIPGuard guard(p.device());
+  long n = p.numel();
+  auto st = cur_stream(p);
+  if (p.scalar_type() == at::kBFloat16) {
+    launch_adam_bf16(bf16_ptr(p), bf16_ptr(g),
+                     has_master ? master.data_ptr<float>() : nullptr,
+                     m.data_ptr<float>(), v.data_ptr<float>(), (float)alpha,
+                     (float)b1, (float)b2, (float)eps, n, has_master, st);
+  } else {
+    launch_adam_f32(p.data_ptr<float>(), g.data_ptr<float>(),
+                    p.data_ptr<float>(), m.data_ptr<float>(),
+                    v.data_ptr<float>(), (float)alpha, (float)b1, (float)b2,
+                    (float)eps, n, false, st);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// softmax-xent (C11)
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> softmax_xent_fwd(at::Tensor logits,
+                                         at::Tensor labels) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2);
+  TORCH_CHECK(labels.scalar_type() == at::kLong);
+  auto lc = logits.contiguous();
+  auto lab = labels.contiguous();
+  long N = lc.size(0);
+  int C = (int)lc.size(1);
+  c10::hip::OptionalHIPGuard guard(lc.device());
+  auto loss = at::empty({N}, lc.options().dtype(at::kFloat));
+  auto dlogits = at::empty_like(lc);
+  auto st = cur_stream(lc);
+  if (lc.scalar_type() == at::kBFloat16) {
+    launch_softmax_xent_bf16(bf16_ptr(lc), lab.data_ptr<long>(),
+                             loss.data_ptr<float>(), bf16_ptr(dlogits), N, C,
+                             st);
+  } else {
+    TORCH_CHECK(lc.scalar_type() == at::kFloat);
+    launch_softmax_xent_f32(lc.data_ptr<float>(), lab.data_ptr<long>(),
+                            loss.data_ptr<float>(), dlogits.data_ptr<float>(),
+                            N, C, st);
+  }
+  return {loss, dlogits};
+}
+
+// ---------------------------------------------------------------------------
+// fused BN(+ReLU) (C9)
+// ---------------------------------------------------------------------------
+
+namespace {
+
+void bn_shapes(const at::Tensor& x, long* M, int* C) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn: NCHW 4-D GPU tensor");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn: channels_last layout required");
+  *C = (int)x.size(1);
+  *M = x.numel() / *C;
+}
+
+void bn_stats(const at::Tensor& x, long M, int C, at::Tensor& sum,
+              at::Tensor& sumsq) {
+  auto st = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16)
+    launch_bn_stats_bf16(bf16_ptr(x), sum.data_ptr<float>(),
+                         sumsq.data_ptr<float>(), M, C, st);
+  else
+    launch_bn_stats_f32(x.data_ptr<float>(), sum.data_ptr<float>(),
+                        sumsq.data_ptr<float>(), M, C, st);
+}
+
+at::Tensor bn_apply(const at::Tensor& x, const at::Tensor& scale,
+                    const at::Tensor& shift, long M, int C, bool relu) {
+  auto y = at::empty_like(x);
+  auto st = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16)
+    launch_bn_apply_bf16(bf16_ptr(x), bf16_ptr(y), scale.data_ptr<float>(),
+                         shift.data_ptr<float>(), M * C, C, relu, st);
+  else
+    launch_bn_apply_f32(x.data_ptr<float>(), y.data_ptr<float>(),
+                        scale.data_ptr<float>(), shift.data_ptr<float>(),
+                        M * C, C, relu, st);
+  return y;
+}
+
+}  // namespace
+
+std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
+                                          at::Tensor b, at::Tensor rm,
+                                          at::Tensor rv, double momentum,
+                                          double eps, bool relu) {
+  long M;
+  int C;
+  bn_shapes(x, &M, &C);
+  c10::hip::OptionalHIPGuard guard(x.device());
+  auto opts = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({C}, opts);
+  auto sumsq = at::zeros({C}, opts);
+  bn_stats(x, M, C, sum, sumsq);
+  auto mean = sum / (double)M;
+  auto var = (sumsq / (double)M - mean * mean).clamp_min(0);
+  auto invstd = at::rsqrt(var + eps);
+  auto scale = w.to(at::kFloat) * invstd;
+  auto shift = b.to(at::kFloat) - mean * scale;
+  {
+    at::NoGradGuard ng;
+    double unbias = M > 1 ? (double)M / (M - 1) : 1.0;
+    rm.mul_(1 - momentum).add_((mean * momentum).to(rm.scalar_type()));
+    rv.mul_(1 - momentum).add_((var * (momentum * unbias)).to(rv.scalar_type()));
+  }
+  auto y = bn_apply(x, scale.contiguous(), shift.contiguous(), M, C, relu);
+  return {y, mean.contiguous(), invstd.contiguous()};
+}
+
+at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
+                            at::Tensor rm, at::Tensor rv, double eps,
+                            bool relu) {
+  long M;
+  int C;
+  bn_shapes(x, &M, &C);
+  c10::hip::OptionalHIPGuard guard(x.device());
+  auto invstd = at::rsqrt(rv.to(at::kFloat) + eps);
+  auto scale = w.to(at::kFloat) * invstd;
+  auto shift = b.to(at::kFloat) - rm.to(at::kFloat) * scale;
+  return bn_apply(x, scale.contiguous(), shift.contiguous(), M, C, relu);
+}
+
+std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
+                                    at::Tensor w, at::Tensor mean,
+                                    at::Tensor invstd, bool relu) {
+  long M;
+  int C;
+  bn_shapes(x, &M, &C);
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
+  c10::hip::OptionalHIPGuard guard(x.device());
+  auto opts = x.options().dtype(at::kFloat);
+  auto s1 = at::zeros({C}, opts);
+  auto s2 = at::zeros({C}, opts);
+  auto st = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16)
+    launch_bn_bwd_reduce_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
+                              mean.data_ptr<float>(),
+                              invstd.data_ptr<float>(), s1.data_ptr<float>(),
+                              s2.data_ptr<float>(), M, C, relu, st);
+  else
+    launch_bn_bwd_reduce_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
+                             y.data_ptr<float>(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), s1.data_ptr<float>(),
+                             s2.data_ptr<float>(), M, C, relu, st);
+  auto a = w.to(at::kFloat) * invstd;  // gamma * invstd
+  auto A = a.contiguous();
+  auto B = (-a * (s2 / (double)M) * invstd).contiguous();
+  auto D = (-a * (s1 / (double)M)
+            + a * (s2 / (double)M) * mean * invstd).contiguous();
+  auto dx = at::empty_like(x);
+  if (x.scalar_type() == at::kBFloat16)
+    launch_bn_bwd_apply_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
+                             bf16_ptr(dx), A.data_ptr<float>(),
+                             B.data_ptr<float>(), D.data_ptr<float>(), M * C,
+                             C, relu, st);
+  else
+    launch_bn_bwd_apply_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
+                            y.data_ptr<float>(), dx.data_ptr<float>(),
+                            A.data_ptr<float>(), B.data_ptr<float>(),
+                            D.data_ptr<float>(), M * C, C, relu, st);
+  return {dx, s2.to(w.scalar_type()), s1.to(w.scalar_type())};
+}
+
+// ---------------------------------------------------------------------------
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "rocps hand-written CDNA4 (gfx950) kernels";
+  m.def("fused_sgd", &fused_sgd);
+  m.def("fused_momentum", &fused_momentum);
+  m.def("fused_adam", &fused_adam);
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("bn_relu_fwd_train", &bn_relu_fwd_train);
+  m.def("bn_relu_fwd_eval", &bn_relu_fwd_eval);
+  m.def("bn_relu_bwd", &bn_relu_bwd);
+}

@@ -1,0 +1,62 @@
+// Declarations of the HIP kernel launchers (defined in *.hip TUs).
+#pragma once
+
+// Host-compilable: only the HIP C API (g++ cannot parse the device
+// headers), bf16 pointers cross the ABI as raw unsigned short*.
+#include <hip/hip_runtime_api.h>
+
+using rbf16 = unsigned short;
+
+// fused_optimizers.hip
+void launch_sgd_f32(float* p, const float* g, float* m, float lr, long n,
+                    bool has_master, hipStream_t stream);
+void launch_sgd_bf16(rbf16* p, const rbf16* g, float* m, float lr, long n,
+                     bool has_master, hipStream_t stream);
+void launch_momentum_f32(float* p, const float* g, float* m, float* v,
+                         float lr, float mu, bool nesterov, long n,
+                         bool has_master, hipStream_t stream);
+void launch_momentum_bf16(rbf16* p, const rbf16* g, float* m, float* v,
+                          float lr, float mu, bool nesterov, long n,
+                          bool has_master, hipStream_t stream);
+void launch_adam_f32(float* p, const float* g, float* mst, float* m, float* v,
+                     float alpha, float b1, float b2, float eps, long n,
+                     bool has_master, hipStream_t stream);
+void launch_adam_bf16(rbf16* p, const rbf16* g, float* mst, float* m,
+                      float* v, float alpha, float b1, float b2, float eps,
+                      long n, bool has_master, hipStream_t stream);
+
+// softmax_xent.hip
+void launch_softmax_xent_f32(const float* logits, const long* labels,
+                             float* loss, float* dlogits, long N, int C,
+                             hipStream_t stream);
+void launch_softmax_xent_bf16(const rbf16* logits, const long* labels,
+                              float* loss, rbf16* dlogits, long N, int C,
+                              hipStream_t stream);
+
+// bn_relu.hip
+void launch_bn_stats_f32(const float* x, float* s, float* ss, long M, int C,
+                         hipStream_t st);
+void launch_bn_stats_bf16(const rbf16* x, float* s, float* ss, long M, int C,
+                          hipStream_t st);
+void launch_bn_apply_f32(const float* x, float* y, const float* sc,
+                         const float* sh, long total, int C, bool relu,
+                         hipStream_t st);
+void launch_bn_apply_bf16(const rbf16* x, rbf16* y, const float* sc,
+                          const float* sh, long total, int C, bool relu,
+                          hipStream_t st);
+void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
+                              const float* mean, const float* invstd,
+                              float* s1, float* s2, long M, int C, bool relu,
+                              hipStream_t st);
+void launch_bn_bwd_reduce_bf16(const rbf16* x, const rbf16* dy,
+                               const rbf16* y, const float* mean,
+                               const float* invstd, float* s1, float* s2,
+                               long M, int C, bool relu, hipStream_t st);
+void launch_bn_bwd_apply_f32(const float* x, const float* dy, const float* y,
+                             float* dx, const float* A, const float* B,
+                             const float* D, long total, int C, bool relu,
+                             hipStream_t st);
+void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy,
+                              const rbf16* y, rbf16* dx, const float* A,
+                              const float* B, const float* D, long total,
+                              int C, bool relu, hipStream_t st);

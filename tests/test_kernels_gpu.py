@@ -1,0 +1,225 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32
+reference (the CPU path of the same op). Run on an MI355X with
+``pytest -m gpu``."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():  # collected on CPU boxes, all skipped
+    pytest.skip("needs an MI355X", allow_module_level=True)
+
+DEV = "cuda:0"
+
+
+def test_hip_extension_present():
+    """On a GPU box the native extension MUST be loaded — a silent
+    eager fallback would invalidate every GPU result."""
+    from rocps.ops import hip_available, hip_ext
+    assert hip_available(), "rocps._hipops missing on a GPU machine"
+    assert hip_ext().__file__.startswith("/")
+
+
+# ---------------------------------------------------------------------------
+# fused optimizers (C8)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_sgd(dtype):
+    from rocps.train.optimizer import GradientDescentOptimizer
+    torch.manual_seed(0)
+    n = 64 * 1000 + 64
+    p32 = torch.randn(n)
+    g32 = torch.randn(n)
+    p = p32.to(dtype).to(DEV)
+    opt = GradientDescentOptimizer(0.1)
+    opt.apply_flat(p, g32.to(dtype).to(DEV), slot="s")
+    torch.cuda.synchronize()
+    if dtype == torch.float32:
+        ref = p32 - 0.1 * g32
+        assert torch.allclose(p.cpu(), ref, atol=1e-6)
+    else:
+        ref = p32.to(dtype).float() - 0.1 * g32.to(dtype).float()
+        master = opt._state["s"]["master"].cpu()
+        assert torch.allclose(master, ref, atol=1e-6)
+        assert torch.equal(p.cpu(), master.bfloat16())
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("nesterov", [False, True])
+def test_fused_momentum(dtype, nesterov):
+    from rocps.train.optimizer import MomentumOptimizer
+    torch.manual_seed(1)
+    n = 4096 + 8  # unaligned tail exercised
+    p0 = torch.randn(n)
+    p = p0.to(dtype).to(DEV)
+    opt = MomentumOptimizer(0.05, 0.9, use_nesterov=nesterov)
+    gs = [torch.randn(n) for _ in range(4)]
+    for g in gs:
+        opt.apply_flat(p, g.to(dtype).to(DEV), slot="s")
+    torch.cuda.synchronize()
+    # fp32 reference with the same start values (post-dtype-rounding)
+    m = p0.to(dtype).float()
+    v = torch.zeros(n)
+    for g in gs:
+        g32 = g.to(dtype).float()
+        v = 0.9 * v + g32
+        m = m - 0.05 * (g32 + 0.9 * v) if nesterov else m - 0.05 * v
+    got = (opt._state["s"]["master"].cpu() if dtype == torch.bfloat16
+           else p.cpu())
+    assert torch.allclose(got, m, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_adam(dtype):
+    from rocps.train.optimizer import AdamOptimizer
+    torch.manual_seed(2)
+    n = 10000
+    p0 = torch.randn(n)
+    p = p0.to(dtype).to(DEV)
+    opt = AdamOptimizer(1e-2, 0.9, 0.999, 1e-8)
+    gs = [torch.randn(n) for _ in range(5)]
+    for g in gs:
+        opt.begin_step()
+        opt.apply_flat(p, g.to(dtype).to(DEV), slot="s")
+    torch.cuda.synchronize()
+    m = torch.zeros(n)
+    v = torch.zeros(n)
+    ref = p0.to(dtype).float()
+    for t, g in enumerate(gs, start=1):
+        g32 = g.to(dtype).float()
+        m = 0.9 * m + 0.1 * g32
+        v = 0.999 * v + 0.001 * g32 * g32
+        alpha = 1e-2 * (1 - 0.999 ** t) ** 0.5 / (1 - 0.9 ** t)
+        ref = ref - alpha * m / (v.sqrt() + 1e-8)
+    got = (opt._state["s"]["master"].cpu() if dtype == torch.bfloat16
+           else p.cpu())
+    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# softmax-xent (C11)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("shape", [(128, 10), (64, 1000), (8, 30522),
+                                   (33, 17)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_softmax_xent(shape, dtype):
+    from rocps.ops.softmax_xent import softmax_cross_entropy_with_logits
+    torch.manual_seed(3)
+    N, C = shape
+    logits_cpu = (torch.randn(N, C) * 3).to(dtype)
+    labels = torch.randint(0, C, (N,))
+    # GPU fused
+    lg = logits_cpu.to(DEV).requires_grad_(True)
+    loss_g = softmax_cross_entropy_with_logits(lg, labels.to(DEV))
+    loss_g.sum().backward()
+    torch.cuda.synchronize()
+    # CPU fp32 reference (same op's CPU path)
+    lc = logits_cpu.clone().requires_grad_(True)
+    loss_c = softmax_cross_entropy_with_logits(lc, labels)
+    loss_c.sum().backward()
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    assert torch.allclose(loss_g.cpu(), loss_c, atol=tol, rtol=tol)
+    assert torch.allclose(lg.grad.float().cpu(), lc.grad.float(),
+                          atol=tol, rtol=tol)
+
+
+# ---------------------------------------------------------------------------
+# fused BN+ReLU (C9)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("relu", [True, False])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape", [(8, 64, 14, 14), (4, 200, 7, 7)])
+def test_bn_relu_train_fwd_bwd(relu, dtype, shape):
+    from rocps.ops.norm import fused_bn_act
+    torch.manual_seed(4)
+    N, C, H, W = shape
+    x_cpu = torch.randn(N, C, H, W).to(dtype)
+    w = torch.rand(C) + 0.5
+    b = torch.randn(C) * 0.1
+    tol = 3e-2 if dtype == torch.bfloat16 else 2e-4
+
+    # GPU fused path
+    xg = (x_cpu.to(DEV).to(memory_format=torch.channels_last)
+          .requires_grad_(True))
+    wg = w.clone().to(DEV).requires_grad_(True)
+    bg = b.clone().to(DEV).requires_grad_(True)
+    rm_g = torch.zeros(C, device=DEV)
+    rv_g = torch.ones(C, device=DEV)
+    yg = fused_bn_act(xg, wg, bg, rm_g, rv_g, training=True, relu=relu)
+    gout = torch.randn_like(yg)
+    yg.backward(gout)
+    torch.cuda.synchronize()
+
+    # CPU fp32 reference (same autograd.Function CPU branch)
+    xc = x_cpu.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    rm_c = torch.zeros(C)
+    rv_c = torch.ones(C)
+    yc = fused_bn_act(xc, wc, bc, rm_c, rv_c, training=True, relu=relu)
+    yc.backward(gout.cpu().contiguous())
+
+    assert torch.allclose(yg.float().cpu(), yc.float(), atol=tol, rtol=tol)
+    assert torch.allclose(rm_g.cpu(), rm_c, atol=tol, rtol=tol)
+    assert torch.allclose(rv_g.cpu(), rv_c, atol=tol, rtol=tol)
+    assert torch.allclose(xg.grad.float().cpu(), xc.grad.float(),
+                          atol=tol, rtol=tol)
+    scale = max(1.0, float(wc.grad.abs().max()))
+    assert torch.allclose(wg.grad.cpu() / scale, wc.grad / scale, atol=5e-3)
+    assert torch.allclose(bg.grad.cpu(), bc.grad, atol=5e-3, rtol=1e-3)
+
+
+def test_bn_relu_eval():
+    from rocps.ops.norm import fused_bn_act
+    torch.manual_seed(5)
+    x = torch.randn(4, 32, 8, 8)
+    w, b = torch.rand(32) + 0.5, torch.randn(32) * 0.1
+    rm, rv = torch.randn(32) * 0.2, torch.rand(32) + 0.5
+    yg = fused_bn_act(x.to(DEV).to(memory_format=torch.channels_last),
+                      w.to(DEV), b.to(DEV), rm.clone().to(DEV),
+                      rv.clone().to(DEV), training=False, relu=True)
+    yc = fused_bn_act(x, w, b, rm.clone(), rv.clone(), training=False,
+                      relu=True)
+    assert torch.allclose(yg.cpu(), yc, atol=1e-5, rtol=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# end-to-end on GPU
+# ---------------------------------------------------------------------------
+
+def test_mnist_engine_gpu_trajectory():
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.train import Trainer, GradientDescentOptimizer
+    torch.manual_seed(0)
+    m_gpu = mnist_mlp(hidden=64)
+    m_cpu = mnist_mlp(hidden=64)
+    m_cpu.load_state_dict(m_gpu.state_dict())
+    tr_g = Trainer(m_gpu, GradientDescentOptimizer(0.1),
+                   device=torch.device(DEV))
+    tr_c = Trainer(m_cpu, GradientDescentOptimizer(0.1),
+                   device=torch.device("cpu"))
+    g = torch.Generator().manual_seed(7)
+    for _ in range(5):
+        x, y = synthetic_mnist_batch(64, generator=g)
+        l_g = tr_g.train_step(lambda: tr_g.model.loss(x.to(DEV), y.to(DEV)))
+        l_c = tr_c.train_step(lambda: tr_c.model.loss(x, y))
+        assert abs(l_g.item() - l_c.item()) < 1e-3
+
+
+def test_resnet50_smoke_bf16():
+    from rocps.models.resnet import resnet50
+    torch.manual_seed(0)
+    m = resnet50().to(DEV).to(memory_format=torch.channels_last).bfloat16()
+    x = torch.randn(8, 3, 224, 224, device=DEV).bfloat16().to(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (8,), device=DEV)
+    loss = m.loss(x, y)
+    loss.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss).all()
+    g = next(m.parameters()).grad
+    assert g is not None and torch.isfinite(g.float()).all()
