@@ -16,7 +16,11 @@ import json
 import os
 import time
 
-import torch
+# fast MIOpen algorithm selection on fresh boxes (no persisted find-db);
+# override with MIOPEN_FIND_MODE=1 for exhaustive tuning runs.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
+import torch  # noqa: E402
 
 
 def parse_args():
