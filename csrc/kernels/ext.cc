@@ -6,15 +6,16 @@
 
 #include <torch/extension.h>
 
-#include <c10/hip/HIPGuard.h>
-#include <c10/hip/HIPStream.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "launchers.h"
 
 namespace {
 
 hipStream_t cur_stream(const at::Tensor& t) {
-  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(
+      t.device().index()).stream();
 }
 
 void check_flat(const at::Tensor& t, const char* name) {
@@ -37,7 +38,7 @@ void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor master, double lr,
   check_flat(p, "params");
   check_flat(g, "grads");
   TORCH_CHECK(p.numel() == g.numel());
-  c10::hip::OptionalHIPGuard guard(p.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(p.device());
   long n = p.numel();
   auto st = cur_stream(p);
   if (p.scalar_type() == at::kBFloat16) {
@@ -60,7 +61,7 @@ void fused_momentum(at::Tensor p, at::Tensor g, at::Tensor master,
   check_flat(mom, "momentum");
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == mom.numel());
   TORCH_CHECK(mom.scalar_type() == at::kFloat);
-  c10::hip::OptionalHIPGuard guard(p.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(p.device());
   long n = p.numel();
   auto st = cur_stream(p);
   if (p.scalar_type() == at::kBFloat16) {
@@ -83,7 +84,7 @@ void fused_adam(at::Tensor p, at::Tensor g, at::Tensor master, at::Tensor m,
   TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel()
               && p.numel() == v.numel());
-  c10::hip::OptionalHIPGuard guard(p.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(p.device());
   long n = p.numel();
   auto st = cur_stream(p);
   if (p.scalar_type() == at::kBFloat16) {
@@ -111,7 +112,7 @@ std::vector<at::Tensor> softmax_xent_fwd(at::Tensor logits,
   auto lab = labels.contiguous();
   long N = lc.size(0);
   int C = (int)lc.size(1);
-  c10::hip::OptionalHIPGuard guard(lc.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(lc.device());
   auto loss = at::empty({N}, lc.options().dtype(at::kFloat));
   auto dlogits = at::empty_like(lc);
   auto st = cur_stream(lc);
@@ -176,7 +177,7 @@ std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
   long M;
   int C;
   bn_shapes(x, &M, &C);
-  c10::hip::OptionalHIPGuard guard(x.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
   auto sum = at::zeros({C}, opts);
   auto sumsq = at::zeros({C}, opts);
@@ -202,7 +203,7 @@ at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
   long M;
   int C;
   bn_shapes(x, &M, &C);
-  c10::hip::OptionalHIPGuard guard(x.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto invstd = at::rsqrt(rv.to(at::kFloat) + eps);
   auto scale = w.to(at::kFloat) * invstd;
   auto shift = b.to(at::kFloat) - rm.to(at::kFloat) * scale;
@@ -216,7 +217,7 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   int C;
   bn_shapes(x, &M, &C);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
-  c10::hip::OptionalHIPGuard guard(x.device());
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
   auto s1 = at::zeros({C}, opts);
   auto s2 = at::zeros({C}, opts);
