@@ -16,11 +16,11 @@ import json
 import os
 import time
 
-# fast MIOpen algorithm selection on fresh boxes (no persisted find-db);
-# override with MIOPEN_FIND_MODE=1 for exhaustive tuning runs.
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
-
 import torch  # noqa: E402
+
+# MIOpen exhaustive find on the conv configs (cached after warmup) —
+# the fallback immediate-mode kernels are ~25x slower on some wrw convs.
+torch.backends.cudnn.benchmark = True
 
 
 def parse_args():
