@@ -24,20 +24,47 @@
 namespace {
 
 // ---- per-channel reduction over rows (stats + bwd reduce) ----------------
+//
+// A 256-thread block is a (CT channels × RT=256/CT rows) tile so small-C
+// layers (the C=64 stem has M=1.6M rows) still put ~256k threads on the
+// chip. Per-thread register accumulation over a grid-strided row range,
+// LDS tree-reduce across the RT row-threads of each channel, then one
+// atomicAdd per channel per block.
+
 template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x,
                                 float* __restrict__ gsum,
-                                float* __restrict__ gsumsq, long M, int C) {
-  int c = blockIdx.y * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+                                float* __restrict__ gsumsq, long M, int C,
+                                int CT) {
+  __shared__ float lsum[256];
+  __shared__ float lss[256];
+  int RT = blockDim.x / CT;
+  int cl = threadIdx.x % CT;
+  int rl = threadIdx.x / CT;
+  int c = blockIdx.y * CT + cl;
   float s = 0.f, ss = 0.f;
-  for (long r = blockIdx.x; r < M; r += gridDim.x) {
-    float v = to_f32<T>(x[r * C + c]);
-    s += v;
-    ss = fmaf(v, v, ss);
+  if (c < C) {
+    long rstride = (long)gridDim.x * RT;
+    for (long r = (long)blockIdx.x * RT + rl; r < M; r += rstride) {
+      float v = to_f32<T>(x[r * C + c]);
+      s += v;
+      ss = fmaf(v, v, ss);
+    }
   }
-  atomicAdd(&gsum[c], s);
-  atomicAdd(&gsumsq[c], ss);
+  lsum[threadIdx.x] = s;
+  lss[threadIdx.x] = ss;
+  __syncthreads();
+  for (int step = (RT >> 1) * CT; step >= CT; step >>= 1) {
+    if (threadIdx.x < step) {
+      lsum[threadIdx.x] += lsum[threadIdx.x + step];
+      lss[threadIdx.x] += lss[threadIdx.x + step];
+    }
+    __syncthreads();
+  }
+  if (rl == 0 && c < C) {
+    atomicAdd(&gsum[c], lsum[cl]);
+    atomicAdd(&gsumsq[c], lss[cl]);
+  }
 }
 
 template <typename T, bool RELU>
@@ -48,21 +75,40 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ gsum_dy,
                                      float* __restrict__ gsum_dyxh,
-                                     long M, int C) {
-  int c = blockIdx.y * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float mu = mean[c], is = invstd[c];
+                                     long M, int C, int CT) {
+  __shared__ float l1[256];
+  __shared__ float l2[256];
+  int RT = blockDim.x / CT;
+  int cl = threadIdx.x % CT;
+  int rl = threadIdx.x / CT;
+  int c = blockIdx.y * CT + cl;
   float s1 = 0.f, s2 = 0.f;
-  for (long r = blockIdx.x; r < M; r += gridDim.x) {
-    long i = r * C + c;
-    float d = to_f32<T>(dy[i]);
-    if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
-    float xh = (to_f32<T>(x[i]) - mu) * is;
-    s1 += d;
-    s2 = fmaf(d, xh, s2);
+  if (c < C) {
+    float mu = mean[c], is = invstd[c];
+    long rstride = (long)gridDim.x * RT;
+    for (long r = (long)blockIdx.x * RT + rl; r < M; r += rstride) {
+      long i = r * C + c;
+      float d = to_f32<T>(dy[i]);
+      if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
+      float xh = (to_f32<T>(x[i]) - mu) * is;
+      s1 += d;
+      s2 = fmaf(d, xh, s2);
+    }
   }
-  atomicAdd(&gsum_dy[c], s1);
-  atomicAdd(&gsum_dyxh[c], s2);
+  l1[threadIdx.x] = s1;
+  l2[threadIdx.x] = s2;
+  __syncthreads();
+  for (int step = (RT >> 1) * CT; step >= CT; step >>= 1) {
+    if (threadIdx.x < step) {
+      l1[threadIdx.x] += l1[threadIdx.x + step];
+      l2[threadIdx.x] += l2[threadIdx.x + step];
+    }
+    __syncthreads();
+  }
+  if (rl == 0 && c < C) {
+    atomicAdd(&gsum_dy[c], l1[cl]);
+    atomicAdd(&gsum_dyxh[c], l2[cl]);
+  }
 }
 
 // ---- elementwise applies --------------------------------------------------
@@ -123,14 +169,26 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
 }  // namespace
 
 // ---- launchers ------------------------------------------------------------
+static int _ct_for(int C) {
+  int ct = 8;
+  while (ct < C && ct < 256) ct <<= 1;
+  return ct;
+}
+
+static dim3 _reduce_grid(long M, int C, int CT) {
+  int RT = 256 / CT;
+  long rb = (M + RT - 1) / RT;
+  if (rb > 1024) rb = 1024;
+  if (rb < 1) rb = 1;
+  return dim3((unsigned)rb, (unsigned)((C + CT - 1) / CT));
+}
+
 template <typename T>
 static void _stats(const T* x, float* gsum, float* gsumsq, long M, int C,
                    hipStream_t stream) {
-  int block = 256;
-  dim3 grid((unsigned)((M + 255) / 256 < 64 ? ((M + 255) / 256 ?: 1) : 64),
-            (C + block - 1) / block);
-  hipLaunchKernelGGL((bn_stats_kernel<T>), grid, dim3(block), 0, stream, x,
-                     gsum, gsumsq, M, C);
+  int CT = _ct_for(C);
+  hipLaunchKernelGGL((bn_stats_kernel<T>), _reduce_grid(M, C, CT), dim3(256),
+                     0, stream, x, gsum, gsumsq, M, C, CT);
 }
 void launch_bn_stats_f32(const float* x, float* s, float* ss, long M, int C,
                          hipStream_t st) { _stats(x, s, ss, M, C, st); }
@@ -173,15 +231,14 @@ static void _bwd_reduce(const T* x, const T* dy, const T* y,
                         const float* mean, const float* invstd, float* s1,
                         float* s2, long M, int C, bool relu,
                         hipStream_t stream) {
-  int block = 256;
-  dim3 grid((unsigned)((M + 255) / 256 < 64 ? ((M + 255) / 256 ?: 1) : 64),
-            (C + block - 1) / block);
+  int CT = _ct_for(C);
+  dim3 grid = _reduce_grid(M, C, CT);
   if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid, dim3(block), 0,
-                       stream, x, dy, y, mean, invstd, s1, s2, M, C);
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid, dim3(256), 0,
+                       stream, x, dy, y, mean, invstd, s1, s2, M, C, CT);
   else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid, dim3(block), 0,
-                       stream, x, dy, y, mean, invstd, s1, s2, M, C);
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid, dim3(256), 0,
+                       stream, x, dy, y, mean, invstd, s1, s2, M, C, CT);
 }
 void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
                               const float* mean, const float* invstd,
