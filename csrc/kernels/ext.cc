@@ -252,6 +252,67 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
 }
 
 // ---------------------------------------------------------------------------
+// fused LayerNorm (bert_ops LN slice)
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
+                               at::Tensor beta, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(gamma.scalar_type() == at::kFloat
+              && beta.scalar_type() == at::kFloat);
+  long N = x.size(0);
+  int D = (int)x.size(1);
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
+  auto y = at::empty_like(x);
+  auto mean = at::empty({N}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty_like(mean);
+  auto st = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16)
+    launch_ln_fwd_bf16(bf16_ptr(x), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), bf16_ptr(y),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(), N, D,
+                       (float)eps, st);
+  else
+    launch_ln_fwd_f32(x.data_ptr<float>(), gamma.data_ptr<float>(),
+                      beta.data_ptr<float>(), y.data_ptr<float>(),
+                      mean.data_ptr<float>(), invstd.data_ptr<float>(), N, D,
+                      (float)eps, st);
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
+                               at::Tensor mean, at::Tensor invstd) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  auto dyc = dy.contiguous();
+  long N = x.size(0);
+  int D = (int)x.size(1);
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
+  auto dx = at::empty_like(x);
+  auto dgamma = at::zeros({D}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({D}, x.options().dtype(at::kFloat));
+  auto st = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16) {
+    launch_ln_bwd_dx_bf16(bf16_ptr(x), bf16_ptr(dyc),
+                          gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                          invstd.data_ptr<float>(), bf16_ptr(dx), N, D, st);
+    launch_ln_bwd_dgb_bf16(bf16_ptr(x), bf16_ptr(dyc),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                           N, D, st);
+  } else {
+    launch_ln_bwd_dx_f32(x.data_ptr<float>(), dyc.data_ptr<float>(),
+                         gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), dx.data_ptr<float>(), N, D,
+                         st);
+    launch_ln_bwd_dgb_f32(x.data_ptr<float>(), dyc.data_ptr<float>(),
+                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                          dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                          N, D, st);
+  }
+  return {dx, dgamma, dbeta};
+}
+
+// ---------------------------------------------------------------------------
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "rocps hand-written CDNA4 (gfx950) kernels";
@@ -262,4 +323,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_relu_fwd_train", &bn_relu_fwd_train);
   m.def("bn_relu_fwd_eval", &bn_relu_fwd_eval);
   m.def("bn_relu_bwd", &bn_relu_bwd);
+  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_bwd", &ln_bwd);
 }

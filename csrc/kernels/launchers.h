@@ -60,3 +60,25 @@ void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy,
                               const rbf16* y, rbf16* dx, const float* A,
                               const float* B, const float* D, long total,
                               int C, bool relu, hipStream_t st);
+
+// layernorm.hip
+void launch_ln_fwd_f32(const float* x, const float* gamma, const float* beta,
+                       float* y, float* mean, float* invstd, long N, int D,
+                       float eps, hipStream_t st);
+void launch_ln_fwd_bf16(const rbf16* x, const float* gamma, const float* beta,
+                        rbf16* y, float* mean, float* invstd, long N, int D,
+                        float eps, hipStream_t st);
+void launch_ln_bwd_dx_f32(const float* x, const float* dy, const float* gamma,
+                          const float* mean, const float* invstd, float* dx,
+                          long N, int D, hipStream_t st);
+void launch_ln_bwd_dx_bf16(const rbf16* x, const rbf16* dy,
+                           const float* gamma, const float* mean,
+                           const float* invstd, rbf16* dx, long N, int D,
+                           hipStream_t st);
+void launch_ln_bwd_dgb_f32(const float* x, const float* dy, const float* mean,
+                           const float* invstd, float* dgamma, float* dbeta,
+                           long N, int D, hipStream_t st);
+void launch_ln_bwd_dgb_bf16(const rbf16* x, const rbf16* dy,
+                            const float* mean, const float* invstd,
+                            float* dgamma, float* dbeta, long N, int D,
+                            hipStream_t st);

@@ -223,3 +223,50 @@ def test_resnet50_smoke_bf16():
     assert torch.isfinite(loss).all()
     g = next(m.parameters()).grad
     assert g is not None and torch.isfinite(g.float()).all()
+
+
+# ---------------------------------------------------------------------------
+# fused LayerNorm (bert_ops)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("D", [768, 320, 1024])
+def test_layernorm_gpu(dtype, D):
+    from rocps.ops.layer_norm import layer_norm
+    torch.manual_seed(6)
+    N = 64
+    x_cpu = torch.randn(N, D).to(dtype)
+    w = (torch.rand(D) + 0.5)
+    b = torch.randn(D) * 0.1
+    tol = 3e-2 if dtype == torch.bfloat16 else 1e-4
+
+    xg = x_cpu.to(DEV).requires_grad_(True)
+    wg = w.clone().to(DEV).requires_grad_(True)
+    bg = b.clone().to(DEV).requires_grad_(True)
+    yg = layer_norm(xg, wg, bg, 1e-12)
+    g = torch.randn(N, D)
+    yg.backward(g.to(DEV).to(dtype))
+    torch.cuda.synchronize()
+
+    xc = x_cpu.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    yc = layer_norm(xc, wc, bc, 1e-12)
+    yc.backward(g.to(dtype))
+    assert torch.allclose(yg.float().cpu(), yc.float(), atol=tol, rtol=tol)
+    assert torch.allclose(xg.grad.float().cpu(), xc.grad.float(), atol=tol,
+                          rtol=tol)
+    assert torch.allclose(wg.grad.cpu(), wc.grad, atol=0.05, rtol=1e-3)
+    assert torch.allclose(bg.grad.cpu(), bc.grad, atol=0.05, rtol=1e-3)
+
+
+def test_bert_tiny_gpu_smoke():
+    from rocps.models.bert import bert_base
+    torch.manual_seed(2)
+    m = bert_base(layers=2).to(DEV).bfloat16()
+    ids = torch.randint(0, 30522, (2, 128), device=DEV)
+    lbl = torch.randint(0, 30522, (2, 128), device=DEV)
+    loss = m.loss(ids, lbl)
+    loss.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
