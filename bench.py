@@ -40,6 +40,8 @@ def parse_args():
     ap.add_argument("--dtype", type=str, default="bf16",
                     choices=["bf16", "fp32"])
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--torch-profile", type=str, default=None,
+                    help="write a torch.profiler summary of the timed steps")
     return ap.parse_args()
 
 
@@ -130,6 +132,12 @@ def main():
     comm.barrier()
     if use_cuda:
         torch.cuda.synchronize()
+    prof = None
+    if args.torch_profile:
+        prof = torch.profiler.profile(
+            activities=[torch.profiler.ProfilerActivity.CPU,
+                        torch.profiler.ProfilerActivity.CUDA])
+        prof.__enter__()
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(args.warmup + i)
@@ -137,6 +145,13 @@ def main():
     if use_cuda:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+    if prof is not None:
+        prof.__exit__(None, None, None)
+        if comm.rank == 0:
+            with open(args.torch_profile, "w") as f:
+                f.write(prof.key_averages().table(
+                    sort_by="self_cuda_time_total" if use_cuda
+                    else "self_cpu_time_total", row_limit=60))
 
     # max over ranks (conservative whole-job time)
     t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
