@@ -169,6 +169,71 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
 }  // namespace
 
 // ---- launchers ------------------------------------------------------------
+// ---- per-channel finalize kernels (C threads) -----------------------------
+// Fold the tiny per-channel math into ONE launch instead of ~25 C-sized
+// torch ops per BN call — that host overhead made ResNet-50 CPU-bound
+// (profiles/r50_b256_torch_profile_hostbound_before.txt).
+
+template <typename P>
+__global__ void bn_finalize_train_kernel(
+    const float* __restrict__ ws,  // [2C]: sum, sumsq
+    const P* __restrict__ w, const P* __restrict__ b, P* __restrict__ rm,
+    P* __restrict__ rv, float momentum, float eps, long M,
+    float* __restrict__ out,  // [4C]: mean, invstd, scale, shift
+    int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean = ws[c] / M;
+  float var = fmaxf(ws[C + c] / M - mean * mean, 0.f);
+  float invstd = rsqrtf(var + eps);
+  float scale = to_f32<P>(w[c]) * invstd;
+  float shift = to_f32<P>(b[c]) - mean * scale;
+  out[c] = mean;
+  out[C + c] = invstd;
+  out[2 * C + c] = scale;
+  out[3 * C + c] = shift;
+  float unbias = M > 1 ? (float)M / (M - 1) : 1.f;
+  rm[c] = from_f32<P>(fmaf(1.f - momentum, to_f32<P>(rm[c]),
+                           momentum * mean));
+  rv[c] = from_f32<P>(fmaf(1.f - momentum, to_f32<P>(rv[c]),
+                           momentum * var * unbias));
+}
+
+template <typename P>
+__global__ void bn_finalize_eval_kernel(const P* __restrict__ w,
+                                        const P* __restrict__ b,
+                                        const P* __restrict__ rm,
+                                        const P* __restrict__ rv, float eps,
+                                        float* __restrict__ out,  // [2C]
+                                        int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float invstd = rsqrtf(to_f32<P>(rv[c]) + eps);
+  float scale = to_f32<P>(w[c]) * invstd;
+  out[c] = scale;
+  out[C + c] = to_f32<P>(b[c]) - to_f32<P>(rm[c]) * scale;
+}
+
+template <typename P>
+__global__ void bn_bwd_finalize_kernel(
+    const float* __restrict__ s12,  // [2C]: sum_dy, sum_dy_xhat
+    const P* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ invstd, long M,
+    float* __restrict__ abd,  // [3C]: A, B, D
+    P* __restrict__ dgamma, P* __restrict__ dbeta, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s1 = s12[c], s2 = s12[C + c];
+  float a = to_f32<P>(w[c]) * invstd[c];
+  float B = -a * (s2 / M) * invstd[c];
+  float D = -a * (s1 / M) + a * (s2 / M) * mean[c] * invstd[c];
+  abd[c] = a;
+  abd[C + c] = B;
+  abd[2 * C + c] = D;
+  dgamma[c] = from_f32<P>(s2);
+  dbeta[c] = from_f32<P>(s1);
+}
+
 static int _ct_for(int C) {
   int ct = 8;
   while (ct < C && ct < 256) ct <<= 1;
@@ -283,4 +348,56 @@ void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy,
              reinterpret_cast<const bf16*>(dy),
              reinterpret_cast<const bf16*>(y), reinterpret_cast<bf16*>(dx), A,
              B, D, total, C, relu, st);
+}
+
+// ---- finalize launchers ---------------------------------------------------
+#define FINALIZE_GRID(C) dim3((unsigned)(((C) + 255) / 256)), dim3(256)
+
+void launch_bn_finalize_train_f32(const float* ws, const float* w,
+                                  const float* b, float* rm, float* rv,
+                                  float momentum, float eps, long M,
+                                  float* out, int C, hipStream_t st) {
+  hipLaunchKernelGGL((bn_finalize_train_kernel<float>), FINALIZE_GRID(C), 0,
+                     st, ws, w, b, rm, rv, momentum, eps, M, out, C);
+}
+void launch_bn_finalize_train_bf16(const float* ws, const rbf16* w,
+                                   const rbf16* b, rbf16* rm, rbf16* rv,
+                                   float momentum, float eps, long M,
+                                   float* out, int C, hipStream_t st) {
+  hipLaunchKernelGGL((bn_finalize_train_kernel<bf16>), FINALIZE_GRID(C), 0,
+                     st, ws, reinterpret_cast<const bf16*>(w),
+                     reinterpret_cast<const bf16*>(b),
+                     reinterpret_cast<bf16*>(rm), reinterpret_cast<bf16*>(rv),
+                     momentum, eps, M, out, C);
+}
+void launch_bn_finalize_eval_f32(const float* w, const float* b,
+                                 const float* rm, const float* rv, float eps,
+                                 float* out, int C, hipStream_t st) {
+  hipLaunchKernelGGL((bn_finalize_eval_kernel<float>), FINALIZE_GRID(C), 0,
+                     st, w, b, rm, rv, eps, out, C);
+}
+void launch_bn_finalize_eval_bf16(const rbf16* w, const rbf16* b,
+                                  const rbf16* rm, const rbf16* rv, float eps,
+                                  float* out, int C, hipStream_t st) {
+  hipLaunchKernelGGL((bn_finalize_eval_kernel<bf16>), FINALIZE_GRID(C), 0,
+                     st, reinterpret_cast<const bf16*>(w),
+                     reinterpret_cast<const bf16*>(b),
+                     reinterpret_cast<const bf16*>(rm),
+                     reinterpret_cast<const bf16*>(rv), eps, out, C);
+}
+void launch_bn_bwd_finalize_f32(const float* s12, const float* w,
+                                const float* mean, const float* invstd,
+                                long M, float* abd, float* dgamma,
+                                float* dbeta, int C, hipStream_t st) {
+  hipLaunchKernelGGL((bn_bwd_finalize_kernel<float>), FINALIZE_GRID(C), 0, st,
+                     s12, w, mean, invstd, M, abd, dgamma, dbeta, C);
+}
+void launch_bn_bwd_finalize_bf16(const float* s12, const rbf16* w,
+                                 const float* mean, const float* invstd,
+                                 long M, float* abd, rbf16* dgamma,
+                                 rbf16* dbeta, int C, hipStream_t st) {
+  hipLaunchKernelGGL((bn_bwd_finalize_kernel<bf16>), FINALIZE_GRID(C), 0, st,
+                     s12, reinterpret_cast<const bf16*>(w), mean, invstd, M,
+                     abd, reinterpret_cast<bf16*>(dgamma),
+                     reinterpret_cast<bf16*>(dbeta), C);
 }

@@ -143,29 +143,36 @@ void bn_shapes(const at::Tensor& x, long* M, int* C) {
   *M = x.numel() / *C;
 }
 
-void bn_stats(const at::Tensor& x, long M, int C, at::Tensor& sum,
-              at::Tensor& sumsq) {
+void bn_stats(const at::Tensor& x, long M, int C, float* sum,
+              float* sumsq) {
   auto st = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16)
-    launch_bn_stats_bf16(bf16_ptr(x), sum.data_ptr<float>(),
-                         sumsq.data_ptr<float>(), M, C, st);
+    launch_bn_stats_bf16(bf16_ptr(x), sum, sumsq, M, C, st);
   else
-    launch_bn_stats_f32(x.data_ptr<float>(), sum.data_ptr<float>(),
-                        sumsq.data_ptr<float>(), M, C, st);
+    launch_bn_stats_f32(x.data_ptr<float>(), sum, sumsq, M, C, st);
 }
 
-at::Tensor bn_apply(const at::Tensor& x, const at::Tensor& scale,
-                    const at::Tensor& shift, long M, int C, bool relu) {
+at::Tensor bn_apply(const at::Tensor& x, const float* scale,
+                    const float* shift, long M, int C, bool relu) {
   auto y = at::empty_like(x);
   auto st = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16)
-    launch_bn_apply_bf16(bf16_ptr(x), bf16_ptr(y), scale.data_ptr<float>(),
-                         shift.data_ptr<float>(), M * C, C, relu, st);
+    launch_bn_apply_bf16(bf16_ptr(x), bf16_ptr(y), scale, shift, M * C, C,
+                         relu, st);
   else
-    launch_bn_apply_f32(x.data_ptr<float>(), y.data_ptr<float>(),
-                        scale.data_ptr<float>(), shift.data_ptr<float>(),
-                        M * C, C, relu, st);
+    launch_bn_apply_f32(x.data_ptr<float>(), y.data_ptr<float>(), scale,
+                        shift, M * C, C, relu, st);
   return y;
+}
+
+void check_bn_params(const at::Tensor& w, const at::Tensor& b,
+                     const at::Tensor& rm, const at::Tensor& rv) {
+  TORCH_CHECK(w.scalar_type() == b.scalar_type()
+                  && w.scalar_type() == rm.scalar_type()
+                  && w.scalar_type() == rv.scalar_type(),
+              "bn: weight/bias/running stats must share a dtype");
+  TORCH_CHECK(w.is_contiguous() && b.is_contiguous() && rm.is_contiguous()
+              && rv.is_contiguous());
 }
 
 }  // namespace
@@ -177,24 +184,26 @@ std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
   long M;
   int C;
   bn_shapes(x, &M, &C);
+  check_bn_params(w, b, rm, rv);
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({C}, opts);
-  auto sumsq = at::zeros({C}, opts);
-  bn_stats(x, M, C, sum, sumsq);
-  auto mean = sum / (double)M;
-  auto var = (sumsq / (double)M - mean * mean).clamp_min(0);
-  auto invstd = at::rsqrt(var + eps);
-  auto scale = w.to(at::kFloat) * invstd;
-  auto shift = b.to(at::kFloat) - mean * scale;
-  {
-    at::NoGradGuard ng;
-    double unbias = M > 1 ? (double)M / (M - 1) : 1.0;
-    rm.mul_(1 - momentum).add_((mean * momentum).to(rm.scalar_type()));
-    rv.mul_(1 - momentum).add_((var * (momentum * unbias)).to(rv.scalar_type()));
-  }
-  auto y = bn_apply(x, scale.contiguous(), shift.contiguous(), M, C, relu);
-  return {y, mean.contiguous(), invstd.contiguous()};
+  auto ws = at::zeros({2 * C}, opts);       // sum | sumsq
+  auto out = at::empty({4 * C}, opts);      // mean | invstd | scale | shift
+  bn_stats(x, M, C, ws.data_ptr<float>(), ws.data_ptr<float>() + C);
+  auto st = cur_stream(x);
+  if (w.scalar_type() == at::kBFloat16)
+    launch_bn_finalize_train_bf16(ws.data_ptr<float>(), bf16_ptr(w),
+                                  bf16_ptr(b), bf16_ptr(rm), bf16_ptr(rv),
+                                  (float)momentum, (float)eps, M,
+                                  out.data_ptr<float>(), C, st);
+  else
+    launch_bn_finalize_train_f32(ws.data_ptr<float>(), w.data_ptr<float>(),
+                                 b.data_ptr<float>(), rm.data_ptr<float>(),
+                                 rv.data_ptr<float>(), (float)momentum,
+                                 (float)eps, M, out.data_ptr<float>(), C, st);
+  auto y = bn_apply(x, out.data_ptr<float>() + 2 * C,
+                    out.data_ptr<float>() + 3 * C, M, C, relu);
+  return {y, out.narrow(0, 0, C), out.narrow(0, C, C)};
 }
 
 at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
@@ -203,11 +212,20 @@ at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
   long M;
   int C;
   bn_shapes(x, &M, &C);
+  check_bn_params(w, b, rm, rv);
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
-  auto invstd = at::rsqrt(rv.to(at::kFloat) + eps);
-  auto scale = w.to(at::kFloat) * invstd;
-  auto shift = b.to(at::kFloat) - rm.to(at::kFloat) * scale;
-  return bn_apply(x, scale.contiguous(), shift.contiguous(), M, C, relu);
+  auto out = at::empty({2 * C}, x.options().dtype(at::kFloat));
+  auto st = cur_stream(x);
+  if (w.scalar_type() == at::kBFloat16)
+    launch_bn_finalize_eval_bf16(bf16_ptr(w), bf16_ptr(b), bf16_ptr(rm),
+                                 bf16_ptr(rv), (float)eps,
+                                 out.data_ptr<float>(), C, st);
+  else
+    launch_bn_finalize_eval_f32(w.data_ptr<float>(), b.data_ptr<float>(),
+                                rm.data_ptr<float>(), rv.data_ptr<float>(),
+                                (float)eps, out.data_ptr<float>(), C, st);
+  return bn_apply(x, out.data_ptr<float>(), out.data_ptr<float>() + C, M, C,
+                  relu);
 }
 
 std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
@@ -217,38 +235,50 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   int C;
   bn_shapes(x, &M, &C);
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(mean.is_contiguous() && invstd.is_contiguous());
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
-  auto s1 = at::zeros({C}, opts);
-  auto s2 = at::zeros({C}, opts);
+  auto s12 = at::zeros({2 * C}, opts);
+  auto abd = at::empty({3 * C}, opts);
+  auto dgb = at::empty({2 * C}, x.options().dtype(w.scalar_type()));
   auto st = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16)
     launch_bn_bwd_reduce_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
                               mean.data_ptr<float>(),
-                              invstd.data_ptr<float>(), s1.data_ptr<float>(),
-                              s2.data_ptr<float>(), M, C, relu, st);
+                              invstd.data_ptr<float>(),
+                              s12.data_ptr<float>(),
+                              s12.data_ptr<float>() + C, M, C, relu, st);
   else
     launch_bn_bwd_reduce_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
                              y.data_ptr<float>(), mean.data_ptr<float>(),
-                             invstd.data_ptr<float>(), s1.data_ptr<float>(),
-                             s2.data_ptr<float>(), M, C, relu, st);
-  auto a = w.to(at::kFloat) * invstd;  // gamma * invstd
-  auto A = a.contiguous();
-  auto B = (-a * (s2 / (double)M) * invstd).contiguous();
-  auto D = (-a * (s1 / (double)M)
-            + a * (s2 / (double)M) * mean * invstd).contiguous();
+                             invstd.data_ptr<float>(), s12.data_ptr<float>(),
+                             s12.data_ptr<float>() + C, M, C, relu, st);
+  if (w.scalar_type() == at::kBFloat16)
+    launch_bn_bwd_finalize_bf16(s12.data_ptr<float>(), bf16_ptr(w),
+                                mean.data_ptr<float>(),
+                                invstd.data_ptr<float>(), M,
+                                abd.data_ptr<float>(), bf16_ptr(dgb),
+                                bf16_ptr(dgb) + C, C, st);
+  else
+    launch_bn_bwd_finalize_f32(s12.data_ptr<float>(), w.data_ptr<float>(),
+                               mean.data_ptr<float>(),
+                               invstd.data_ptr<float>(), M,
+                               abd.data_ptr<float>(), dgb.data_ptr<float>(),
+                               dgb.data_ptr<float>() + C, C, st);
   auto dx = at::empty_like(x);
   if (x.scalar_type() == at::kBFloat16)
     launch_bn_bwd_apply_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
-                             bf16_ptr(dx), A.data_ptr<float>(),
-                             B.data_ptr<float>(), D.data_ptr<float>(), M * C,
-                             C, relu, st);
+                             bf16_ptr(dx), abd.data_ptr<float>(),
+                             abd.data_ptr<float>() + C,
+                             abd.data_ptr<float>() + 2 * C, M * C, C, relu,
+                             st);
   else
     launch_bn_bwd_apply_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
                             y.data_ptr<float>(), dx.data_ptr<float>(),
-                            A.data_ptr<float>(), B.data_ptr<float>(),
-                            D.data_ptr<float>(), M * C, C, relu, st);
-  return {dx, s2.to(w.scalar_type()), s1.to(w.scalar_type())};
+                            abd.data_ptr<float>(), abd.data_ptr<float>() + C,
+                            abd.data_ptr<float>() + 2 * C, M * C, C, relu,
+                            st);
+  return {dx, dgb.narrow(0, 0, C), dgb.narrow(0, C, C)};
 }
 
 // ---------------------------------------------------------------------------

@@ -82,3 +82,27 @@ void launch_ln_bwd_dgb_bf16(const rbf16* x, const rbf16* dy,
                             const float* mean, const float* invstd,
                             float* dgamma, float* dbeta, long N, int D,
                             hipStream_t st);
+
+// bn_relu.hip finalize kernels
+void launch_bn_finalize_train_f32(const float* ws, const float* w,
+                                  const float* b, float* rm, float* rv,
+                                  float momentum, float eps, long M,
+                                  float* out, int C, hipStream_t st);
+void launch_bn_finalize_train_bf16(const float* ws, const rbf16* w,
+                                   const rbf16* b, rbf16* rm, rbf16* rv,
+                                   float momentum, float eps, long M,
+                                   float* out, int C, hipStream_t st);
+void launch_bn_finalize_eval_f32(const float* w, const float* b,
+                                 const float* rm, const float* rv, float eps,
+                                 float* out, int C, hipStream_t st);
+void launch_bn_finalize_eval_bf16(const rbf16* w, const rbf16* b,
+                                  const rbf16* rm, const rbf16* rv, float eps,
+                                  float* out, int C, hipStream_t st);
+void launch_bn_bwd_finalize_f32(const float* s12, const float* w,
+                                const float* mean, const float* invstd,
+                                long M, float* abd, float* dgamma,
+                                float* dbeta, int C, hipStream_t st);
+void launch_bn_bwd_finalize_bf16(const float* s12, const rbf16* w,
+                                 const float* mean, const float* invstd,
+                                 long M, float* abd, rbf16* dgamma,
+                                 rbf16* dbeta, int C, hipStream_t st);
