@@ -1,178 +1,344 @@
-// Fused BatchNorm(+ReLU) forward/backward on NHWC — C9's BN slice.
+// Fused BatchNorm(+residual-add)(+ReLU) forward/backward on NHWC — C9.
 //
-// Reference behavior: TF FusedBatchNormV3[Grad] + Relu[Grad]
+// Reference behavior: TF FusedBatchNormV3[Grad] + AddV2 + Relu[Grad]
 // (tensorflow/core/kernels/fused_batch_norm_op.cc, relu_op.cc) on every
-// conv output of ResNet/W-ResNet. Recast for CDNA4 NHWC:
+// conv output of ResNet/W-ResNet. CDNA4 NHWC design:
 //
-//   fwd train: stats kernel (per-channel sum/sumsq, fp32, per-thread
-//     register accumulation over a row range + 2 device atomicAdds per
-//     thread) → per-channel scale/shift folded on the small C-sized
-//     tensors → apply kernel y = relu(x*scale[c]+shift[c]) vectorized
-//     8 channels/lane (bf16 ushort8 = 16 B, Guideline 13).
-//   bwd: reduce kernel (sum_dy, sum_dy*xhat with the ReLU mask folded)
-//     → per-channel constants → apply kernel dx = A[c]*dy_m + B[c]*x +
-//     D[c]. dgamma/dbeta fall out of the reduction.
+// - Column-resident threads: each thread owns an 8-channel group
+//   (16 B of bf16 = the coalescing sweet spot, Guideline 13), loads its
+//   per-channel constants ONCE into registers, and walks rows — no
+//   per-element modulo, no per-element constant reloads.
+// - stats / bwd-reduce: a 256-thread block = G col-groups × RT=256/G
+//   row lanes; per-thread 8-wide register accumulators, LDS tree-reduce
+//   over row lanes, then 16 atomicAdds per col-group per block.
+// - apply: y = relu(x*scale[c] + shift[c] [+ res]); the bottleneck's
+//   residual add + ReLU fuse into the bn3 epilogue (ADD variant), and
+//   the backward emits d_residual = relu-masked dy in the same pass.
+// - per-channel finalize kernels keep ALL small math on device (one
+//   launch, C threads) — C-sized host torch ops made the step CPU-bound
+//   (profiles/r50_b256_torch_profile_hostbound_before.txt).
 //
-// Layout contract: x is channels_last → memory is (M=N*H*W) rows × C
-// contiguous channels. Lanes map to adjacent channels (coalesced); the
-// per-channel constant arrays are C-sized and L2-resident.
+// Layout contract: channels_last → memory (M=N*H*W) rows × C contiguous.
+// The vector path requires C % 8 == 0 (every ResNet/BERT channel count);
+// a scalar fallback covers odd C.
 
 #include "common.h"
 #include "launchers.h"
 
 namespace {
 
-// ---- per-channel reduction over rows (stats + bwd reduce) ----------------
-//
-// A 256-thread block is a (CT channels × RT=256/CT rows) tile so small-C
-// layers (the C=64 stem has M=1.6M rows) still put ~256k threads on the
-// chip. Per-thread register accumulation over a grid-strided row range,
-// LDS tree-reduce across the RT row-threads of each channel, then one
-// atomicAdd per channel per block.
+struct alignas(16) u16x8v { unsigned short v[8]; };
+struct alignas(16) f32x4v { float v[4]; };
 
 template <typename T>
+__device__ __forceinline__ void load8v(const T* p, float* out);
+template <>
+__device__ __forceinline__ void load8v<bf16>(const bf16* p, float* out) {
+  u16x8v raw = *reinterpret_cast<const u16x8v*>(p);
+#pragma unroll
+  for (int k = 0; k < 8; k++)
+    out[k] = __uint_as_float((unsigned int)raw.v[k] << 16);
+}
+template <>
+__device__ __forceinline__ void load8v<float>(const float* p, float* out) {
+  f32x4v a = *reinterpret_cast<const f32x4v*>(p);
+  f32x4v b = *reinterpret_cast<const f32x4v*>(p + 4);
+#pragma unroll
+  for (int k = 0; k < 4; k++) { out[k] = a.v[k]; out[4 + k] = b.v[k]; }
+}
+
+template <typename T>
+__device__ __forceinline__ void store8v(T* p, const float* in);
+template <>
+__device__ __forceinline__ void store8v<bf16>(bf16* p, const float* in) {
+  u16x8v raw;
+#pragma unroll
+  for (int k = 0; k < 8; k++)
+    raw.v[k] = __hip_bfloat16_raw(__float2bfloat16(in[k])).x;
+  *reinterpret_cast<u16x8v*>(p) = raw;
+}
+template <>
+__device__ __forceinline__ void store8v<float>(float* p, const float* in) {
+  f32x4v a, b;
+#pragma unroll
+  for (int k = 0; k < 4; k++) { a.v[k] = in[k]; b.v[k] = in[4 + k]; }
+  *reinterpret_cast<f32x4v*>(p) = a;
+  *reinterpret_cast<f32x4v*>(p + 4) = b;
+}
+
+// ---------------------------------------------------------------------------
+// stats: ws[0:C] += sum(x), ws[C:2C] += sum(x^2)
+// ---------------------------------------------------------------------------
+template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x,
-                                float* __restrict__ gsum,
-                                float* __restrict__ gsumsq, long M, int C,
-                                int CT) {
-  __shared__ float lsum[256];
-  __shared__ float lss[256];
-  int RT = blockDim.x / CT;
-  int cl = threadIdx.x % CT;
-  int rl = threadIdx.x / CT;
-  int c = blockIdx.y * CT + cl;
-  float s = 0.f, ss = 0.f;
-  if (c < C) {
+                                float* __restrict__ ws, long M, int C,
+                                int G) {
+  __shared__ float lds[256 * 16];
+  int cols = C >> 3;
+  int RT = blockDim.x / G;
+  int gl = threadIdx.x % G;
+  int rl = threadIdx.x / G;
+  int cg = blockIdx.y * G + gl;
+  float s[8] = {0.f}, ss[8] = {0.f};
+  if (cg < cols) {
     long rstride = (long)gridDim.x * RT;
+    const T* base = x + cg * 8;
     for (long r = (long)blockIdx.x * RT + rl; r < M; r += rstride) {
-      float v = to_f32<T>(x[r * C + c]);
-      s += v;
-      ss = fmaf(v, v, ss);
+      float v[8];
+      load8v<T>(base + r * C, v);
+#pragma unroll
+      for (int k = 0; k < 8; k++) {
+        s[k] += v[k];
+        ss[k] = fmaf(v[k], v[k], ss[k]);
+      }
     }
   }
-  lsum[threadIdx.x] = s;
-  lss[threadIdx.x] = ss;
+  float* mine = &lds[threadIdx.x * 16];
+#pragma unroll
+  for (int k = 0; k < 8; k++) { mine[k] = s[k]; mine[8 + k] = ss[k]; }
   __syncthreads();
-  for (int step = (RT >> 1) * CT; step >= CT; step >>= 1) {
-    if (threadIdx.x < step) {
-      lsum[threadIdx.x] += lsum[threadIdx.x + step];
-      lss[threadIdx.x] += lss[threadIdx.x + step];
+  for (int step = (RT >> 1) * G; step >= G; step >>= 1) {
+    if ((int)threadIdx.x < step) {
+      float* other = &lds[(threadIdx.x + step) * 16];
+#pragma unroll
+      for (int k = 0; k < 16; k++) mine[k] += other[k];
     }
     __syncthreads();
   }
-  if (rl == 0 && c < C) {
-    atomicAdd(&gsum[c], lsum[cl]);
-    atomicAdd(&gsumsq[c], lss[cl]);
+  if (rl == 0 && cg < cols) {
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      atomicAdd(&ws[cg * 8 + k], mine[k]);
+      atomicAdd(&ws[C + cg * 8 + k], mine[8 + k]);
+    }
   }
 }
 
+// scalar fallback (odd C)
+template <typename T>
+__global__ void bn_stats_scalar_kernel(const T* __restrict__ x,
+                                       float* __restrict__ ws, long M,
+                                       int C) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, ss = 0.f;
+  for (long r = blockIdx.x; r < M; r += gridDim.x) {
+    float v = to_f32<T>(x[r * C + c]);
+    s += v;
+    ss = fmaf(v, v, ss);
+  }
+  atomicAdd(&ws[c], s);
+  atomicAdd(&ws[C + c], ss);
+}
+
+// ---------------------------------------------------------------------------
+// bwd reduce: s12[0:C] += sum(dy_m), s12[C:2C] += sum(dy_m * xhat)
+// ---------------------------------------------------------------------------
 template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ dy,
                                      const T* __restrict__ y,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
-                                     float* __restrict__ gsum_dy,
-                                     float* __restrict__ gsum_dyxh,
-                                     long M, int C, int CT) {
-  __shared__ float l1[256];
-  __shared__ float l2[256];
-  int RT = blockDim.x / CT;
-  int cl = threadIdx.x % CT;
-  int rl = threadIdx.x / CT;
-  int c = blockIdx.y * CT + cl;
-  float s1 = 0.f, s2 = 0.f;
-  if (c < C) {
-    float mu = mean[c], is = invstd[c];
+                                     float* __restrict__ s12, long M, int C,
+                                     int G) {
+  __shared__ float lds[256 * 16];
+  int cols = C >> 3;
+  int RT = blockDim.x / G;
+  int gl = threadIdx.x % G;
+  int rl = threadIdx.x / G;
+  int cg = blockIdx.y * G + gl;
+  float s1[8] = {0.f}, s2[8] = {0.f};
+  if (cg < cols) {
+    float mu[8], is[8];
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      mu[k] = mean[cg * 8 + k];
+      is[k] = invstd[cg * 8 + k];
+    }
     long rstride = (long)gridDim.x * RT;
+    long off = cg * 8;
     for (long r = (long)blockIdx.x * RT + rl; r < M; r += rstride) {
-      long i = r * C + c;
-      float d = to_f32<T>(dy[i]);
-      if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
-      float xh = (to_f32<T>(x[i]) - mu) * is;
-      s1 += d;
-      s2 = fmaf(d, xh, s2);
+      float xv[8], dv[8];
+      load8v<T>(x + r * C + off, xv);
+      load8v<T>(dy + r * C + off, dv);
+      if (RELU) {
+        float yv[8];
+        load8v<T>(y + r * C + off, yv);
+#pragma unroll
+        for (int k = 0; k < 8; k++) dv[k] = yv[k] > 0.f ? dv[k] : 0.f;
+      }
+#pragma unroll
+      for (int k = 0; k < 8; k++) {
+        s1[k] += dv[k];
+        s2[k] = fmaf(dv[k], (xv[k] - mu[k]) * is[k], s2[k]);
+      }
     }
   }
-  l1[threadIdx.x] = s1;
-  l2[threadIdx.x] = s2;
+  float* mine = &lds[threadIdx.x * 16];
+#pragma unroll
+  for (int k = 0; k < 8; k++) { mine[k] = s1[k]; mine[8 + k] = s2[k]; }
   __syncthreads();
-  for (int step = (RT >> 1) * CT; step >= CT; step >>= 1) {
-    if (threadIdx.x < step) {
-      l1[threadIdx.x] += l1[threadIdx.x + step];
-      l2[threadIdx.x] += l2[threadIdx.x + step];
+  for (int step = (RT >> 1) * G; step >= G; step >>= 1) {
+    if ((int)threadIdx.x < step) {
+      float* other = &lds[(threadIdx.x + step) * 16];
+#pragma unroll
+      for (int k = 0; k < 16; k++) mine[k] += other[k];
     }
     __syncthreads();
   }
-  if (rl == 0 && c < C) {
-    atomicAdd(&gsum_dy[c], l1[cl]);
-    atomicAdd(&gsum_dyxh[c], l2[cl]);
-  }
-}
-
-// ---- elementwise applies --------------------------------------------------
-struct alignas(16) u16x8 { unsigned short v[8]; };
-
-template <typename T, bool RELU, bool VEC8>
-__global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                const float* __restrict__ scale,
-                                const float* __restrict__ shift,
-                                long total, int C) {
-  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  if (VEC8) {
-    long nvec = total / 8;
-    for (long i = tid; i < nvec; i += stride) {
-      long base = i * 8;
-      int c0 = (int)(base % C);
-      float out[8];
+  if (rl == 0 && cg < cols) {
 #pragma unroll
-      for (int k = 0; k < 8; k++) {
-        float v = to_f32<T>(x[base + k]);
-        int c = c0 + k;  // C % 8 == 0 → the 8 lanes stay inside one row
-        v = fmaf(v, scale[c], shift[c]);
-        out[k] = RELU ? fmaxf(v, 0.f) : v;
-      }
-#pragma unroll
-      for (int k = 0; k < 8; k++) y[base + k] = from_f32<T>(out[k]);
-    }
-  } else {
-    for (long i = tid; i < total; i += stride) {
-      int c = (int)(i % C);
-      float v = fmaf(to_f32<T>(x[i]), scale[c], shift[c]);
-      y[i] = from_f32<T>(RELU ? fmaxf(v, 0.f) : v);
+    for (int k = 0; k < 8; k++) {
+      atomicAdd(&s12[cg * 8 + k], mine[k]);
+      atomicAdd(&s12[C + cg * 8 + k], mine[8 + k]);
     }
   }
 }
 
 template <typename T, bool RELU>
+__global__ void bn_bwd_reduce_scalar_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, const T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ s12, long M, int C) {
+  int c = blockIdx.y * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mu = mean[c], is = invstd[c];
+  float a = 0.f, b = 0.f;
+  for (long r = blockIdx.x; r < M; r += gridDim.x) {
+    long i = r * C + c;
+    float d = to_f32<T>(dy[i]);
+    if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
+    a += d;
+    b = fmaf(d, (to_f32<T>(x[i]) - mu) * is, b);
+  }
+  atomicAdd(&s12[c], a);
+  atomicAdd(&s12[C + c], b);
+}
+
+// ---------------------------------------------------------------------------
+// apply: y = act(x*scale + shift [+ res]); column-resident threads
+// ---------------------------------------------------------------------------
+template <typename T, bool RELU, bool ADD>
+__global__ void bn_apply_kernel(const T* __restrict__ x,
+                                const T* __restrict__ res, T* __restrict__ y,
+                                const float* __restrict__ scale,
+                                const float* __restrict__ shift, long M,
+                                int C) {
+  int cols = C >> 3;
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long nthreads = (long)gridDim.x * blockDim.x;
+  long rstride = nthreads / cols;
+  if (tid >= rstride * cols) return;
+  int cg = (int)(tid % cols);
+  long row0 = tid / cols;
+  long off = (long)cg * 8;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int k = 0; k < 8; k++) {
+    sc[k] = scale[off + k];
+    sh[k] = shift[off + k];
+  }
+  for (long r = row0; r < M; r += rstride) {
+    float v[8];
+    load8v<T>(x + r * C + off, v);
+#pragma unroll
+    for (int k = 0; k < 8; k++) v[k] = fmaf(v[k], sc[k], sh[k]);
+    if (ADD) {
+      float rv[8];
+      load8v<T>(res + r * C + off, rv);
+#pragma unroll
+      for (int k = 0; k < 8; k++) v[k] += rv[k];
+    }
+    if (RELU) {
+#pragma unroll
+      for (int k = 0; k < 8; k++) v[k] = fmaxf(v[k], 0.f);
+    }
+    store8v<T>(y + r * C + off, v);
+  }
+}
+
+template <typename T, bool RELU, bool ADD>
+__global__ void bn_apply_scalar_kernel(const T* __restrict__ x,
+                                       const T* __restrict__ res,
+                                       T* __restrict__ y,
+                                       const float* __restrict__ scale,
+                                       const float* __restrict__ shift,
+                                       long total, int C) {
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = tid; i < total; i += stride) {
+    int c = (int)(i % C);
+    float v = fmaf(to_f32<T>(x[i]), scale[c], shift[c]);
+    if (ADD) v += to_f32<T>(res[i]);
+    y[i] = from_f32<T>(RELU ? fmaxf(v, 0.f) : v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bwd apply: dx = A*dy_m + B*x + D; (ADD) dres = dy_m
+// ---------------------------------------------------------------------------
+template <typename T, bool RELU, bool ADD>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const T* __restrict__ dy,
                                     const T* __restrict__ y,
-                                    T* __restrict__ dx,
-                                    const float* __restrict__ A,
-                                    const float* __restrict__ B,
-                                    const float* __restrict__ D,
-                                    long total, int C) {
+                                    T* __restrict__ dx, T* __restrict__ dres,
+                                    const float* __restrict__ abd, long M,
+                                    int C) {
+  int cols = C >> 3;
+  long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long nthreads = (long)gridDim.x * blockDim.x;
+  long rstride = nthreads / cols;
+  if (tid >= rstride * cols) return;
+  int cg = (int)(tid % cols);
+  long row0 = tid / cols;
+  long off = (long)cg * 8;
+  float A[8], B[8], D[8];
+#pragma unroll
+  for (int k = 0; k < 8; k++) {
+    A[k] = abd[off + k];
+    B[k] = abd[C + off + k];
+    D[k] = abd[2 * C + off + k];
+  }
+  for (long r = row0; r < M; r += rstride) {
+    float xv[8], dv[8];
+    load8v<T>(x + r * C + off, xv);
+    load8v<T>(dy + r * C + off, dv);
+    if (RELU) {
+      float yv[8];
+      load8v<T>(y + r * C + off, yv);
+#pragma unroll
+      for (int k = 0; k < 8; k++) dv[k] = yv[k] > 0.f ? dv[k] : 0.f;
+    }
+    if (ADD) store8v<T>(dres + r * C + off, dv);
+    float o[8];
+#pragma unroll
+    for (int k = 0; k < 8; k++)
+      o[k] = fmaf(A[k], dv[k], fmaf(B[k], xv[k], D[k]));
+    store8v<T>(dx + r * C + off, o);
+  }
+}
+
+template <typename T, bool RELU, bool ADD>
+__global__ void bn_bwd_apply_scalar_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, const T* __restrict__ y,
+    T* __restrict__ dx, T* __restrict__ dres, const float* __restrict__ abd,
+    long total, int C) {
   long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = tid; i < total; i += stride) {
     int c = (int)(i % C);
     float d = to_f32<T>(dy[i]);
     if (RELU) d = to_f32<T>(y[i]) > 0.f ? d : 0.f;
-    float r = fmaf(A[c], d, fmaf(B[c], to_f32<T>(x[i]), D[c]));
-    dx[i] = from_f32<T>(r);
+    if (ADD) dres[i] = from_f32<T>(d);
+    dx[i] = from_f32<T>(fmaf(abd[c], d, fmaf(abd[C + c], to_f32<T>(x[i]),
+                                             abd[2 * C + c])));
   }
 }
 
-}  // namespace
-
-// ---- launchers ------------------------------------------------------------
-// ---- per-channel finalize kernels (C threads) -----------------------------
-// Fold the tiny per-channel math into ONE launch instead of ~25 C-sized
-// torch ops per BN call — that host overhead made ResNet-50 CPU-bound
-// (profiles/r50_b256_torch_profile_hostbound_before.txt).
-
+// ---------------------------------------------------------------------------
+// per-channel finalize kernels (C threads, one launch)
+// ---------------------------------------------------------------------------
 template <typename P>
 __global__ void bn_finalize_train_kernel(
     const float* __restrict__ ws,  // [2C]: sum, sumsq
@@ -224,129 +390,185 @@ __global__ void bn_bwd_finalize_kernel(
   if (c >= C) return;
   float s1 = s12[c], s2 = s12[C + c];
   float a = to_f32<P>(w[c]) * invstd[c];
-  float B = -a * (s2 / M) * invstd[c];
-  float D = -a * (s1 / M) + a * (s2 / M) * mean[c] * invstd[c];
   abd[c] = a;
-  abd[C + c] = B;
-  abd[2 * C + c] = D;
+  abd[C + c] = -a * (s2 / M) * invstd[c];
+  abd[2 * C + c] = -a * (s1 / M) + a * (s2 / M) * mean[c] * invstd[c];
   dgamma[c] = from_f32<P>(s2);
   dbeta[c] = from_f32<P>(s1);
 }
 
-static int _ct_for(int C) {
-  int ct = 8;
-  while (ct < C && ct < 256) ct <<= 1;
-  return ct;
+// ---------------------------------------------------------------------------
+// launch helpers
+// ---------------------------------------------------------------------------
+inline int g_for(int C) {  // col-groups per 256-thread block
+  int cols = C >> 3;
+  return cols < 32 ? cols : 32;
 }
 
-static dim3 _reduce_grid(long M, int C, int CT) {
-  int RT = 256 / CT;
+inline dim3 reduce_grid(long M, int C, int G) {
+  int RT = 256 / G;
   long rb = (M + RT - 1) / RT;
   if (rb > 1024) rb = 1024;
   if (rb < 1) rb = 1;
-  return dim3((unsigned)rb, (unsigned)((C + CT - 1) / CT));
+  int cols = C >> 3;
+  return dim3((unsigned)rb, (unsigned)((cols + G - 1) / G));
 }
 
-template <typename T>
-static void _stats(const T* x, float* gsum, float* gsumsq, long M, int C,
-                   hipStream_t stream) {
-  int CT = _ct_for(C);
-  hipLaunchKernelGGL((bn_stats_kernel<T>), _reduce_grid(M, C, CT), dim3(256),
-                     0, stream, x, gsum, gsumsq, M, C, CT);
-}
-void launch_bn_stats_f32(const float* x, float* s, float* ss, long M, int C,
-                         hipStream_t st) { _stats(x, s, ss, M, C, st); }
-void launch_bn_stats_bf16(const rbf16* x, float* s, float* ss, long M, int C,
-                          hipStream_t st) {
-  _stats(reinterpret_cast<const bf16*>(x), s, ss, M, C, st);
+inline int apply_grid(long M, int C) {
+  long work = M * (C >> 3);
+  long blocks = (work + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
 }
 
+}  // namespace
+
+// ---- exported launchers ---------------------------------------------------
+
 template <typename T>
-static void _apply(const T* x, T* y, const float* scale, const float* shift,
-                   long total, int C, bool relu, hipStream_t stream) {
-  int block = 256;
-  int grid = rocps_grid_for((total + 7) / 8, block);
-  bool vec8 = (C % 8 == 0);
-  if (relu && vec8)
-    hipLaunchKernelGGL((bn_apply_kernel<T, true, true>), dim3(grid),
-                       dim3(block), 0, stream, x, y, scale, shift, total, C);
-  else if (relu)
-    hipLaunchKernelGGL((bn_apply_kernel<T, true, false>), dim3(grid),
-                       dim3(block), 0, stream, x, y, scale, shift, total, C);
-  else if (vec8)
-    hipLaunchKernelGGL((bn_apply_kernel<T, false, true>), dim3(grid),
-                       dim3(block), 0, stream, x, y, scale, shift, total, C);
-  else
-    hipLaunchKernelGGL((bn_apply_kernel<T, false, false>), dim3(grid),
-                       dim3(block), 0, stream, x, y, scale, shift, total, C);
+static void _stats(const T* x, float* ws, long M, int C, hipStream_t stream) {
+  if (C % 8 == 0) {
+    int G = g_for(C);
+    hipLaunchKernelGGL((bn_stats_kernel<T>), reduce_grid(M, C, G), dim3(256),
+                       0, stream, x, ws, M, C, G);
+  } else {
+    dim3 grid((unsigned)(M < 512 ? (M ? M : 1) : 512),
+              (unsigned)((C + 255) / 256));
+    hipLaunchKernelGGL((bn_stats_scalar_kernel<T>), grid, dim3(256), 0,
+                       stream, x, ws, M, C);
+  }
 }
-void launch_bn_apply_f32(const float* x, float* y, const float* sc,
-                         const float* sh, long total, int C, bool relu,
-                         hipStream_t st) { _apply(x, y, sc, sh, total, C, relu, st); }
-void launch_bn_apply_bf16(const rbf16* x, rbf16* y, const float* sc,
-                          const float* sh, long total, int C, bool relu,
+void launch_bn_stats_f32(const float* x, float* ws, long M, int C,
+                         hipStream_t st) { _stats(x, ws, M, C, st); }
+void launch_bn_stats_bf16(const rbf16* x, float* ws, long M, int C,
                           hipStream_t st) {
-  _apply(reinterpret_cast<const bf16*>(x), reinterpret_cast<bf16*>(y), sc, sh,
-         total, C, relu, st);
+  _stats(reinterpret_cast<const bf16*>(x), ws, M, C, st);
 }
 
 template <typename T>
 static void _bwd_reduce(const T* x, const T* dy, const T* y,
-                        const float* mean, const float* invstd, float* s1,
-                        float* s2, long M, int C, bool relu,
-                        hipStream_t stream) {
-  int CT = _ct_for(C);
-  dim3 grid = _reduce_grid(M, C, CT);
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid, dim3(256), 0,
-                       stream, x, dy, y, mean, invstd, s1, s2, M, C, CT);
-  else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid, dim3(256), 0,
-                       stream, x, dy, y, mean, invstd, s1, s2, M, C, CT);
+                        const float* mean, const float* invstd, float* s12,
+                        long M, int C, bool relu, hipStream_t stream) {
+  if (C % 8 == 0) {
+    int G = g_for(C);
+    dim3 grid = reduce_grid(M, C, G);
+    if (relu)
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid, dim3(256), 0,
+                         stream, x, dy, y, mean, invstd, s12, M, C, G);
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid, dim3(256), 0,
+                         stream, x, dy, y, mean, invstd, s12, M, C, G);
+  } else {
+    dim3 grid((unsigned)(M < 512 ? (M ? M : 1) : 512),
+              (unsigned)((C + 255) / 256));
+    if (relu)
+      hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<T, true>), grid,
+                         dim3(256), 0, stream, x, dy, y, mean, invstd, s12, M,
+                         C);
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce_scalar_kernel<T, false>), grid,
+                         dim3(256), 0, stream, x, dy, y, mean, invstd, s12, M,
+                         C);
+  }
 }
 void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
                               const float* mean, const float* invstd,
-                              float* s1, float* s2, long M, int C, bool relu,
+                              float* s12, long M, int C, bool relu,
                               hipStream_t st) {
-  _bwd_reduce(x, dy, y, mean, invstd, s1, s2, M, C, relu, st);
+  _bwd_reduce(x, dy, y, mean, invstd, s12, M, C, relu, st);
 }
 void launch_bn_bwd_reduce_bf16(const rbf16* x, const rbf16* dy,
                                const rbf16* y, const float* mean,
-                               const float* invstd, float* s1, float* s2,
-                               long M, int C, bool relu, hipStream_t st) {
+                               const float* invstd, float* s12, long M, int C,
+                               bool relu, hipStream_t st) {
   _bwd_reduce(reinterpret_cast<const bf16*>(x),
               reinterpret_cast<const bf16*>(dy),
-              reinterpret_cast<const bf16*>(y), mean, invstd, s1, s2, M, C,
-              relu, st);
+              reinterpret_cast<const bf16*>(y), mean, invstd, s12, M, C, relu,
+              st);
 }
 
 template <typename T>
-static void _bwd_apply(const T* x, const T* dy, const T* y, T* dx,
-                       const float* A, const float* B, const float* D,
-                       long total, int C, bool relu, hipStream_t stream) {
-  int block = 256;
-  int grid = rocps_grid_for(total, block);
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, true>), dim3(grid),
-                       dim3(block), 0, stream, x, dy, y, dx, A, B, D, total, C);
-  else
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, false>), dim3(grid),
-                       dim3(block), 0, stream, x, dy, y, dx, A, B, D, total, C);
+static void _apply(const T* x, const T* res, T* y, const float* scale,
+                   const float* shift, long M, int C, bool relu,
+                   hipStream_t stream) {
+  bool add = res != nullptr;
+  if (C % 8 == 0) {
+    int grid = apply_grid(M, C);
+#define AP(R, A)                                                          \
+    hipLaunchKernelGGL((bn_apply_kernel<T, R, A>), dim3(grid), dim3(256), \
+                       0, stream, x, res, y, scale, shift, M, C)
+    if (relu && add) AP(true, true);
+    else if (relu) AP(true, false);
+    else if (add) AP(false, true);
+    else AP(false, false);
+#undef AP
+  } else {
+    int grid = (int)((M * C + 255) / 256 < 2048 ? ((M * C + 255) / 256 ?: 1)
+                                                : 2048);
+#define APS(R, A)                                                           \
+    hipLaunchKernelGGL((bn_apply_scalar_kernel<T, R, A>), dim3(grid),       \
+                       dim3(256), 0, stream, x, res, y, scale, shift, M * C, C)
+    if (relu && add) APS(true, true);
+    else if (relu) APS(true, false);
+    else if (add) APS(false, true);
+    else APS(false, false);
+#undef APS
+  }
+}
+void launch_bn_apply_f32(const float* x, const float* res, float* y,
+                         const float* sc, const float* sh, long M, int C,
+                         bool relu, hipStream_t st) {
+  _apply(x, res, y, sc, sh, M, C, relu, st);
+}
+void launch_bn_apply_bf16(const rbf16* x, const rbf16* res, rbf16* y,
+                          const float* sc, const float* sh, long M, int C,
+                          bool relu, hipStream_t st) {
+  _apply(reinterpret_cast<const bf16*>(x),
+         reinterpret_cast<const bf16*>(res), reinterpret_cast<bf16*>(y), sc,
+         sh, M, C, relu, st);
+}
+
+template <typename T>
+static void _bwd_apply(const T* x, const T* dy, const T* y, T* dx, T* dres,
+                       const float* abd, long M, int C, bool relu,
+                       hipStream_t stream) {
+  bool add = dres != nullptr;
+  if (C % 8 == 0) {
+    int grid = apply_grid(M, C);
+#define BAP(R, A)                                                            \
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, R, A>), dim3(grid),           \
+                       dim3(256), 0, stream, x, dy, y, dx, dres, abd, M, C)
+    if (relu && add) BAP(true, true);
+    else if (relu) BAP(true, false);
+    else if (add) BAP(false, true);
+    else BAP(false, false);
+#undef BAP
+  } else {
+    int grid = (int)((M * C + 255) / 256 < 2048 ? ((M * C + 255) / 256 ?: 1)
+                                                : 2048);
+#define BAPS(R, A)                                                           \
+    hipLaunchKernelGGL((bn_bwd_apply_scalar_kernel<T, R, A>), dim3(grid),    \
+                       dim3(256), 0, stream, x, dy, y, dx, dres, abd, M * C, C)
+    if (relu && add) BAPS(true, true);
+    else if (relu) BAPS(true, false);
+    else if (add) BAPS(false, true);
+    else BAPS(false, false);
+#undef BAPS
+  }
 }
 void launch_bn_bwd_apply_f32(const float* x, const float* dy, const float* y,
-                             float* dx, const float* A, const float* B,
-                             const float* D, long total, int C, bool relu,
-                             hipStream_t st) {
-  _bwd_apply(x, dy, y, dx, A, B, D, total, C, relu, st);
+                             float* dx, float* dres, const float* abd, long M,
+                             int C, bool relu, hipStream_t st) {
+  _bwd_apply(x, dy, y, dx, dres, abd, M, C, relu, st);
 }
-void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy,
-                              const rbf16* y, rbf16* dx, const float* A,
-                              const float* B, const float* D, long total,
-                              int C, bool relu, hipStream_t st) {
+void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy, const rbf16* y,
+                              rbf16* dx, rbf16* dres, const float* abd,
+                              long M, int C, bool relu, hipStream_t st) {
   _bwd_apply(reinterpret_cast<const bf16*>(x),
              reinterpret_cast<const bf16*>(dy),
-             reinterpret_cast<const bf16*>(y), reinterpret_cast<bf16*>(dx), A,
-             B, D, total, C, relu, st);
+             reinterpret_cast<const bf16*>(y), reinterpret_cast<bf16*>(dx),
+             reinterpret_cast<bf16*>(dres), abd, M, C, relu, st);
 }
 
 // ---- finalize launchers ---------------------------------------------------

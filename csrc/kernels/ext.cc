@@ -143,25 +143,26 @@ void bn_shapes(const at::Tensor& x, long* M, int* C) {
   *M = x.numel() / *C;
 }
 
-void bn_stats(const at::Tensor& x, long M, int C, float* sum,
-              float* sumsq) {
+void bn_stats(const at::Tensor& x, long M, int C, float* ws) {
   auto st = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16)
-    launch_bn_stats_bf16(bf16_ptr(x), sum, sumsq, M, C, st);
+    launch_bn_stats_bf16(bf16_ptr(x), ws, M, C, st);
   else
-    launch_bn_stats_f32(x.data_ptr<float>(), sum, sumsq, M, C, st);
+    launch_bn_stats_f32(x.data_ptr<float>(), ws, M, C, st);
 }
 
-at::Tensor bn_apply(const at::Tensor& x, const float* scale,
-                    const float* shift, long M, int C, bool relu) {
+at::Tensor bn_apply(const at::Tensor& x, const at::Tensor* res,
+                    const float* scale, const float* shift, long M, int C,
+                    bool relu) {
   auto y = at::empty_like(x);
   auto st = cur_stream(x);
   if (x.scalar_type() == at::kBFloat16)
-    launch_bn_apply_bf16(bf16_ptr(x), bf16_ptr(y), scale, shift, M * C, C,
-                         relu, st);
+    launch_bn_apply_bf16(bf16_ptr(x), res ? bf16_ptr(*res) : nullptr,
+                         bf16_ptr(y), scale, shift, M, C, relu, st);
   else
-    launch_bn_apply_f32(x.data_ptr<float>(), y.data_ptr<float>(), scale,
-                        shift, M * C, C, relu, st);
+    launch_bn_apply_f32(x.data_ptr<float>(),
+                        res ? res->data_ptr<float>() : nullptr,
+                        y.data_ptr<float>(), scale, shift, M, C, relu, st);
   return y;
 }
 
@@ -180,16 +181,22 @@ void check_bn_params(const at::Tensor& w, const at::Tensor& b,
 std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
                                           at::Tensor b, at::Tensor rm,
                                           at::Tensor rv, double momentum,
-                                          double eps, bool relu) {
+                                          double eps, bool relu,
+                                          std::optional<at::Tensor> res) {
   long M;
   int C;
   bn_shapes(x, &M, &C);
   check_bn_params(w, b, rm, rv);
+  if (res) {
+    TORCH_CHECK(res->is_contiguous(at::MemoryFormat::ChannelsLast)
+                && res->sizes() == x.sizes()
+                && res->scalar_type() == x.scalar_type());
+  }
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
   auto ws = at::zeros({2 * C}, opts);       // sum | sumsq
   auto out = at::empty({4 * C}, opts);      // mean | invstd | scale | shift
-  bn_stats(x, M, C, ws.data_ptr<float>(), ws.data_ptr<float>() + C);
+  bn_stats(x, M, C, ws.data_ptr<float>());
   auto st = cur_stream(x);
   if (w.scalar_type() == at::kBFloat16)
     launch_bn_finalize_train_bf16(ws.data_ptr<float>(), bf16_ptr(w),
@@ -201,14 +208,14 @@ std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
                                  b.data_ptr<float>(), rm.data_ptr<float>(),
                                  rv.data_ptr<float>(), (float)momentum,
                                  (float)eps, M, out.data_ptr<float>(), C, st);
-  auto y = bn_apply(x, out.data_ptr<float>() + 2 * C,
+  auto y = bn_apply(x, res ? &*res : nullptr, out.data_ptr<float>() + 2 * C,
                     out.data_ptr<float>() + 3 * C, M, C, relu);
   return {y, out.narrow(0, 0, C), out.narrow(0, C, C)};
 }
 
 at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
                             at::Tensor rm, at::Tensor rv, double eps,
-                            bool relu) {
+                            bool relu, std::optional<at::Tensor> res) {
   long M;
   int C;
   bn_shapes(x, &M, &C);
@@ -224,13 +231,14 @@ at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
     launch_bn_finalize_eval_f32(w.data_ptr<float>(), b.data_ptr<float>(),
                                 rm.data_ptr<float>(), rv.data_ptr<float>(),
                                 (float)eps, out.data_ptr<float>(), C, st);
-  return bn_apply(x, out.data_ptr<float>(), out.data_ptr<float>() + C, M, C,
-                  relu);
+  return bn_apply(x, res ? &*res : nullptr, out.data_ptr<float>(),
+                  out.data_ptr<float>() + C, M, C, relu);
 }
 
 std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                                     at::Tensor w, at::Tensor mean,
-                                    at::Tensor invstd, bool relu) {
+                                    at::Tensor invstd, bool relu,
+                                    bool has_res) {
   long M;
   int C;
   bn_shapes(x, &M, &C);
@@ -246,13 +254,12 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
     launch_bn_bwd_reduce_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
                               mean.data_ptr<float>(),
                               invstd.data_ptr<float>(),
-                              s12.data_ptr<float>(),
-                              s12.data_ptr<float>() + C, M, C, relu, st);
+                              s12.data_ptr<float>(), M, C, relu, st);
   else
     launch_bn_bwd_reduce_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
                              y.data_ptr<float>(), mean.data_ptr<float>(),
                              invstd.data_ptr<float>(), s12.data_ptr<float>(),
-                             s12.data_ptr<float>() + C, M, C, relu, st);
+                             M, C, relu, st);
   if (w.scalar_type() == at::kBFloat16)
     launch_bn_bwd_finalize_bf16(s12.data_ptr<float>(), bf16_ptr(w),
                                 mean.data_ptr<float>(),
@@ -266,18 +273,20 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                                abd.data_ptr<float>(), dgb.data_ptr<float>(),
                                dgb.data_ptr<float>() + C, C, st);
   auto dx = at::empty_like(x);
+  at::Tensor dres;
+  if (has_res) dres = at::empty_like(x);
   if (x.scalar_type() == at::kBFloat16)
     launch_bn_bwd_apply_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
-                             bf16_ptr(dx), abd.data_ptr<float>(),
-                             abd.data_ptr<float>() + C,
-                             abd.data_ptr<float>() + 2 * C, M * C, C, relu,
-                             st);
+                             bf16_ptr(dx),
+                             has_res ? bf16_ptr(dres) : nullptr,
+                             abd.data_ptr<float>(), M, C, relu, st);
   else
     launch_bn_bwd_apply_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
                             y.data_ptr<float>(), dx.data_ptr<float>(),
-                            abd.data_ptr<float>(), abd.data_ptr<float>() + C,
-                            abd.data_ptr<float>() + 2 * C, M * C, C, relu,
-                            st);
+                            has_res ? dres.data_ptr<float>() : nullptr,
+                            abd.data_ptr<float>(), M, C, relu, st);
+  if (has_res)
+    return {dx, dgb.narrow(0, 0, C), dgb.narrow(0, C, C), dres};
   return {dx, dgb.narrow(0, 0, C), dgb.narrow(0, C, C)};
 }
 

@@ -34,32 +34,30 @@ void launch_softmax_xent_bf16(const rbf16* logits, const long* labels,
                               hipStream_t stream);
 
 // bn_relu.hip
-void launch_bn_stats_f32(const float* x, float* s, float* ss, long M, int C,
+void launch_bn_stats_f32(const float* x, float* ws, long M, int C,
                          hipStream_t st);
-void launch_bn_stats_bf16(const rbf16* x, float* s, float* ss, long M, int C,
+void launch_bn_stats_bf16(const rbf16* x, float* ws, long M, int C,
                           hipStream_t st);
-void launch_bn_apply_f32(const float* x, float* y, const float* sc,
-                         const float* sh, long total, int C, bool relu,
-                         hipStream_t st);
-void launch_bn_apply_bf16(const rbf16* x, rbf16* y, const float* sc,
-                          const float* sh, long total, int C, bool relu,
-                          hipStream_t st);
+void launch_bn_apply_f32(const float* x, const float* res, float* y,
+                         const float* sc, const float* sh, long M, int C,
+                         bool relu, hipStream_t st);
+void launch_bn_apply_bf16(const rbf16* x, const rbf16* res, rbf16* y,
+                          const float* sc, const float* sh, long M, int C,
+                          bool relu, hipStream_t st);
 void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
                               const float* mean, const float* invstd,
-                              float* s1, float* s2, long M, int C, bool relu,
+                              float* s12, long M, int C, bool relu,
                               hipStream_t st);
 void launch_bn_bwd_reduce_bf16(const rbf16* x, const rbf16* dy,
                                const rbf16* y, const float* mean,
-                               const float* invstd, float* s1, float* s2,
-                               long M, int C, bool relu, hipStream_t st);
+                               const float* invstd, float* s12, long M, int C,
+                               bool relu, hipStream_t st);
 void launch_bn_bwd_apply_f32(const float* x, const float* dy, const float* y,
-                             float* dx, const float* A, const float* B,
-                             const float* D, long total, int C, bool relu,
-                             hipStream_t st);
-void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy,
-                              const rbf16* y, rbf16* dx, const float* A,
-                              const float* B, const float* D, long total,
-                              int C, bool relu, hipStream_t st);
+                             float* dx, float* dres, const float* abd, long M,
+                             int C, bool relu, hipStream_t st);
+void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy, const rbf16* y,
+                              rbf16* dx, rbf16* dres, const float* abd,
+                              long M, int C, bool relu, hipStream_t st);
 
 // layernorm.hip
 void launch_ln_fwd_f32(const float* x, const float* gamma, const float* beta,

@@ -46,18 +46,17 @@ class Bottleneck(nn.Module):
         self.conv2 = conv3x3(width, width, stride)
         self.bn2 = BatchNormAct2d(width, act="relu")
         self.conv3 = conv1x1(width, planes * self.expansion)
-        self.bn3 = BatchNormAct2d(planes * self.expansion, act="none")
+        # bn3 fuses the skip-add + final ReLU into its epilogue (C9)
+        self.bn3 = BatchNormAct2d(planes * self.expansion, act="add_relu")
         self.downsample = downsample
-        self.relu = nn.ReLU(inplace=True)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return self.relu(out + identity)
+        return self.bn3(self.conv3(out), residual=identity)
 
 
 class ResNet(nn.Module):

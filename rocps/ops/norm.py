@@ -1,19 +1,19 @@
-"""Fused BatchNorm(+ReLU) — the BN part of capability C9.
+"""Fused BatchNorm(+residual-add)(+ReLU) — the BN slice of C9.
 
-Reference: TF ``FusedBatchNormV3``/``Grad`` + separate ``Relu`` — on
-every conv output of ResNet/W-ResNet. Here both directions run as
+Reference: TF ``FusedBatchNormV3``/``Grad`` + ``AddV2`` + ``Relu`` on
+every conv output of ResNet/W-ResNet. Both directions run as
 hand-written CDNA4 kernels on NHWC bf16/fp32 tensors
 (csrc/kernels/bn_relu.hip):
 
-- fwd: one channel-parallel stats kernel (sum/sumsq, fp32 accum,
-  per-block partials + device atomics) + one vectorized
-  normalize+affine+ReLU kernel → 2 HBM passes instead of 4.
-- bwd: one reduction kernel (sum_dy, sum_dy*xhat with the ReLU mask
-  folded in) + one apply kernel producing dx → 2 passes instead of 5.
+- fwd train: channel-group stats kernel + per-channel finalize (one
+  C-thread launch; keeps all small math off the host) + one vectorized
+  apply ``y = relu(x*scale[c]+shift[c] [+ residual])``.
+- bwd: one masked reduction + finalize + one apply producing dx (and
+  d_residual = relu-masked dy when the bottleneck's skip-add is fused).
 
-CPU path: plain fp32 PyTorch reference (same math, used by numerics
-tests). Training-mode stats are biased variance (TF/torch semantics);
-running stats update uses unbiased variance like torch BatchNorm2d.
+CPU path: plain fp32 PyTorch reference (same math; what the GPU
+numerics tests compare against). Training stats are biased variance;
+running-var update is unbiased, matching torch BatchNorm2d.
 """
 
 from __future__ import annotations
@@ -22,7 +22,6 @@ from typing import Optional
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from rocps.ops import hip_ext, require_hip_for
 
@@ -31,22 +30,27 @@ __all__ = ["BatchNormAct2d", "fused_bn_act"]
 
 class _FusedBNAct(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var,
+    def forward(ctx, x, residual, weight, bias, running_mean, running_var,
                 training: bool, momentum: float, eps: float, relu: bool):
+        has_res = residual is not None
+        ctx.relu = relu
+        ctx.has_res = has_res
         if require_hip_for(x):
             assert x.dim() == 4
             xc = x.contiguous(memory_format=torch.channels_last)
+            res = (residual.contiguous(memory_format=torch.channels_last)
+                   if has_res else None)
             if training:
                 y, save_mean, save_invstd = hip_ext().bn_relu_fwd_train(
                     xc, weight, bias, running_mean, running_var,
-                    momentum, eps, relu)
+                    momentum, eps, relu, res)
             else:
                 y = hip_ext().bn_relu_fwd_eval(
-                    xc, weight, bias, running_mean, running_var, eps, relu)
+                    xc, weight, bias, running_mean, running_var, eps, relu,
+                    res)
                 save_mean = running_mean.float()
                 save_invstd = (running_var.float() + eps).rsqrt()
             ctx.save_for_backward(xc, weight, save_mean, save_invstd, y)
-            ctx.relu = relu
             ctx.hip = True
             return y
         # CPU fp32 reference path
@@ -57,9 +61,11 @@ class _FusedBNAct(torch.autograd.Function):
             var = x32.var(dim=(0, 2, 3), unbiased=False)
             n = x.numel() / x.shape[1]
             with torch.no_grad():
-                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_mean.mul_(1 - momentum).add_(
+                    mean.to(running_mean.dtype), alpha=momentum)
                 running_var.mul_(1 - momentum).add_(
-                    var * n / max(n - 1, 1), alpha=momentum)
+                    (var * n / max(n - 1, 1)).to(running_var.dtype),
+                    alpha=momentum)
         else:
             mean = running_mean.float()
             var = running_var.float()
@@ -67,11 +73,12 @@ class _FusedBNAct(torch.autograd.Function):
         xhat = (x32 - mean[None, :, None, None]) * invstd[None, :, None, None]
         y = xhat * weight.float()[None, :, None, None] \
             + bias.float()[None, :, None, None]
+        if has_res:
+            y = y + residual.float()
         if relu:
             y = y.relu()
         y = y.to(x.dtype)
         ctx.save_for_backward(x, weight, mean, invstd, y)
-        ctx.relu = relu
         return y
 
     @staticmethod
@@ -79,12 +86,16 @@ class _FusedBNAct(torch.autograd.Function):
         x, weight, mean, invstd, y = ctx.saved_tensors
         if ctx.hip:
             dyc = dy.contiguous(memory_format=torch.channels_last)
-            dx, dgamma, dbeta = hip_ext().bn_relu_bwd(
-                x, dyc, y, weight, mean, invstd, ctx.relu)
-            return (dx, dgamma, dbeta, None, None, None, None, None, None)
+            outs = hip_ext().bn_relu_bwd(
+                x, dyc, y, weight, mean, invstd, ctx.relu, ctx.has_res)
+            dx, dgamma, dbeta = outs[0], outs[1], outs[2]
+            dres = outs[3] if ctx.has_res else None
+            return (dx, dres, dgamma, dbeta,
+                    None, None, None, None, None, None)
         dy32 = dy.float()
         if ctx.relu:
             dy32 = dy32 * (y > 0).float()
+        dres = dy32.to(dy.dtype) if ctx.has_res else None
         x32 = x.float()
         M = x.numel() / x.shape[1]
         xhat = (x32 - mean[None, :, None, None]) * invstd[None, :, None, None]
@@ -93,19 +104,21 @@ class _FusedBNAct(torch.autograd.Function):
         g = weight.float()[None, :, None, None] * invstd[None, :, None, None]
         dx = g * (dy32 - dbeta[None, :, None, None] / M
                   - xhat * dgamma[None, :, None, None] / M)
-        return (dx.to(x.dtype), dgamma.to(weight.dtype),
+        return (dx.to(x.dtype), dres, dgamma.to(weight.dtype),
                 dbeta.to(weight.dtype), None, None, None, None, None, None)
 
 
 def fused_bn_act(x, weight, bias, running_mean, running_var,
                  training: bool, momentum: float = 0.1, eps: float = 1e-5,
-                 relu: bool = True) -> torch.Tensor:
-    return _FusedBNAct.apply(x, weight, bias, running_mean, running_var,
-                             training, momentum, eps, relu)
+                 relu: bool = True,
+                 residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    return _FusedBNAct.apply(x, residual, weight, bias, running_mean,
+                             running_var, training, momentum, eps, relu)
 
 
 class BatchNormAct2d(nn.Module):
-    """Drop-in BatchNorm2d with fused optional ReLU epilogue.
+    """Drop-in BatchNorm2d with fused optional residual-add + ReLU
+    epilogue (``act``: "relu", "none", "add_relu").
 
     State-dict layout matches nn.BatchNorm2d (weight/bias/running_mean/
     running_var/num_batches_tracked) so checkpoints interop.
@@ -114,7 +127,7 @@ class BatchNormAct2d(nn.Module):
     def __init__(self, num_features: int, eps: float = 1e-5,
                  momentum: float = 0.1, act: str = "relu"):
         super().__init__()
-        assert act in ("relu", "none")
+        assert act in ("relu", "none", "add_relu")
         self.num_features = num_features
         self.eps = eps
         self.momentum = momentum
@@ -126,12 +139,16 @@ class BatchNormAct2d(nn.Module):
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor,
+                residual: Optional[torch.Tensor] = None) -> torch.Tensor:
         if self.training:
             self.num_batches_tracked += 1
+        if self.act == "add_relu":
+            assert residual is not None, "add_relu needs a residual input"
         return fused_bn_act(x, self.weight, self.bias, self.running_mean,
                             self.running_var, self.training, self.momentum,
-                            self.eps, relu=(self.act == "relu"))
+                            self.eps, relu=(self.act != "none"),
+                            residual=residual)
 
     def extra_repr(self) -> str:
         return f"{self.num_features}, act={self.act}"

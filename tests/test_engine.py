@@ -95,3 +95,18 @@ def test_overlap_launch_order_covers_all_buckets():
     loss.backward()
     eng.finish_backward()
     assert len(eng._launch_order) == len(eng.buckets)
+
+
+def test_resnet_tiny_cpu_smoke():
+    """Bottleneck with fused bn3(add_relu) runs fwd/bwd on CPU."""
+    import torch
+    from rocps.models.resnet import ResNet
+    torch.manual_seed(0)
+    m = ResNet([1, 1, 1, 1], num_classes=10)
+    x = torch.randn(2, 3, 64, 64)
+    y = torch.randint(0, 10, (2,))
+    loss = m.loss(x, y)
+    loss.backward()
+    assert torch.isfinite(loss)
+    for n, p in m.named_parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all(), n

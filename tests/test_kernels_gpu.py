@@ -270,3 +270,43 @@ def test_bert_tiny_gpu_smoke():
     loss.backward()
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_add_relu_fused(dtype):
+    """bn3's fused residual-add + ReLU vs the CPU reference path."""
+    from rocps.ops.norm import fused_bn_act
+    torch.manual_seed(7)
+    N, C, H, W = 4, 64, 8, 8
+    x_cpu = torch.randn(N, C, H, W).to(dtype)
+    r_cpu = torch.randn(N, C, H, W).to(dtype)
+    w = torch.rand(C) + 0.5
+    b = torch.randn(C) * 0.1
+    tol = 3e-2 if dtype == torch.bfloat16 else 2e-4
+
+    xg = (x_cpu.to(DEV).to(memory_format=torch.channels_last)
+          .requires_grad_(True))
+    rg = (r_cpu.to(DEV).to(memory_format=torch.channels_last)
+          .requires_grad_(True))
+    wg = w.clone().to(DEV).requires_grad_(True)
+    bg = b.clone().to(DEV).requires_grad_(True)
+    yg = fused_bn_act(xg, wg, bg, torch.zeros(C, device=DEV),
+                      torch.ones(C, device=DEV), training=True, relu=True,
+                      residual=rg)
+    gout = torch.randn_like(yg)
+    yg.backward(gout)
+    torch.cuda.synchronize()
+
+    xc = x_cpu.clone().requires_grad_(True)
+    rc = r_cpu.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    yc = fused_bn_act(xc, wc, bc, torch.zeros(C), torch.ones(C),
+                      training=True, relu=True, residual=rc)
+    yc.backward(gout.cpu().contiguous())
+    assert torch.allclose(yg.float().cpu(), yc.float(), atol=tol, rtol=tol)
+    assert torch.allclose(xg.grad.float().cpu(), xc.grad.float(), atol=tol,
+                          rtol=tol)
+    assert torch.allclose(rg.grad.float().cpu(), rc.grad.float(), atol=tol,
+                          rtol=tol)
+    assert torch.allclose(bg.grad.cpu(), bc.grad, atol=5e-3, rtol=1e-3)
