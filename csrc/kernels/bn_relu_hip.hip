@@ -1,3 +1,4 @@
+#include "hip/hip_runtime.h"
 // Fused BatchNorm(+residual-add)(+ReLU) forward/backward on NHWC — C9.
 //
 // Reference behavior: TF FusedBatchNormV3[Grad] + AddV2 + Relu[Grad]

@@ -1,0 +1,179 @@
+#include "hip/hip_runtime.h"
+// Fused LayerNorm fwd/bwd — the LN part of `bert_ops` (SURVEY §2d:
+// "fused LN fwd/bwd" for BERT (B·S,768) rows).
+//
+// fwd: one 256-thread block per row; block-reduced mean/var; one read +
+//   one write of the row; mean/invstd saved fp32 for backward.
+// bwd: dx kernel (row-parallel, two block reductions, one fused pass)
+//   + column-reduction kernel for dgamma/dbeta (channel-tiled blocks +
+//   atomics, same scheme as bn_relu.hip).
+
+#include "common.h"
+#include "launchers.h"
+
+namespace {
+
+template <typename T>
+__global__ void ln_fwd_kernel(const T* __restrict__ x,
+                              const float* __restrict__ gamma,
+                              const float* __restrict__ beta,
+                              T* __restrict__ y, float* __restrict__ mean,
+                              float* __restrict__ invstd, int D, float eps) {
+  __shared__ float lds[8];
+  long row = blockIdx.x;
+  const T* xr = x + row * D;
+  T* yr = y + row * D;
+  float s = 0.f, ss = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float v = to_f32<T>(xr[d]);
+    s += v;
+    ss = fmaf(v, v, ss);
+  }
+  s = block_reduce_sum(s, lds);
+  ss = block_reduce_sum(ss, lds);
+  float mu = s / D;
+  float var = fmaxf(ss / D - mu * mu, 0.f);
+  float is = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    mean[row] = mu;
+    invstd[row] = is;
+  }
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float v = (to_f32<T>(xr[d]) - mu) * is;
+    yr[d] = from_f32<T>(fmaf(v, gamma[d], beta[d]));
+  }
+}
+
+template <typename T>
+__global__ void ln_bwd_dx_kernel(const T* __restrict__ x,
+                                 const T* __restrict__ dy,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 T* __restrict__ dx, int D) {
+  __shared__ float lds[8];
+  long row = blockIdx.x;
+  const T* xr = x + row * D;
+  const T* dyr = dy + row * D;
+  T* dxr = dx + row * D;
+  float mu = mean[row], is = invstd[row];
+  float c1 = 0.f, c2 = 0.f;  // mean(dy*g*xhat), mean(dy*g)
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float g = to_f32<T>(dyr[d]) * gamma[d];
+    float xh = (to_f32<T>(xr[d]) - mu) * is;
+    c1 = fmaf(g, xh, c1);
+    c2 += g;
+  }
+  c1 = block_reduce_sum(c1, lds) / D;
+  c2 = block_reduce_sum(c2, lds) / D;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float g = to_f32<T>(dyr[d]) * gamma[d];
+    float xh = (to_f32<T>(xr[d]) - mu) * is;
+    dxr[d] = from_f32<T>(is * (g - c2 - xh * c1));
+  }
+}
+
+// column reduction: dgamma[d] = sum_r dy*xhat, dbeta[d] = sum_r dy
+template <typename T>
+__global__ void ln_bwd_dgb_kernel(const T* __restrict__ x,
+                                  const T* __restrict__ dy,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ invstd,
+                                  float* __restrict__ dgamma,
+                                  float* __restrict__ dbeta, long N, int D,
+                                  int CT) {
+  __shared__ float l1[256];
+  __shared__ float l2[256];
+  int RT = blockDim.x / CT;
+  int cl = threadIdx.x % CT;
+  int rl = threadIdx.x / CT;
+  int d = blockIdx.y * CT + cl;
+  float s1 = 0.f, s2 = 0.f;
+  if (d < D) {
+    long rstride = (long)gridDim.x * RT;
+    for (long r = (long)blockIdx.x * RT + rl; r < N; r += rstride) {
+      float dv = to_f32<T>(dy[r * D + d]);
+      float xh = (to_f32<T>(x[r * D + d]) - mean[r]) * invstd[r];
+      s1 = fmaf(dv, xh, s1);
+      s2 += dv;
+    }
+  }
+  l1[threadIdx.x] = s1;
+  l2[threadIdx.x] = s2;
+  __syncthreads();
+  for (int step = (RT >> 1) * CT; step >= CT; step >>= 1) {
+    if (threadIdx.x < step) {
+      l1[threadIdx.x] += l1[threadIdx.x + step];
+      l2[threadIdx.x] += l2[threadIdx.x + step];
+    }
+    __syncthreads();
+  }
+  if (rl == 0 && d < D) {
+    atomicAdd(&dgamma[d], l1[cl]);
+    atomicAdd(&dbeta[d], l2[cl]);
+  }
+}
+
+static int ln_ct_for(int C) {
+  int ct = 8;
+  while (ct < C && ct < 256) ct <<= 1;
+  return ct;
+}
+
+}  // namespace
+
+void launch_ln_fwd_f32(const float* x, const float* gamma, const float* beta,
+                       float* y, float* mean, float* invstd, long N, int D,
+                       float eps, hipStream_t st) {
+  hipLaunchKernelGGL((ln_fwd_kernel<float>), dim3((unsigned)N), dim3(256), 0,
+                     st, x, gamma, beta, y, mean, invstd, D, eps);
+}
+void launch_ln_fwd_bf16(const rbf16* x, const float* gamma, const float* beta,
+                        rbf16* y, float* mean, float* invstd, long N, int D,
+                        float eps, hipStream_t st) {
+  hipLaunchKernelGGL((ln_fwd_kernel<bf16>), dim3((unsigned)N), dim3(256), 0,
+                     st, reinterpret_cast<const bf16*>(x), gamma, beta,
+                     reinterpret_cast<bf16*>(y), mean, invstd, D, eps);
+}
+
+void launch_ln_bwd_dx_f32(const float* x, const float* dy, const float* gamma,
+                          const float* mean, const float* invstd, float* dx,
+                          long N, int D, hipStream_t st) {
+  hipLaunchKernelGGL((ln_bwd_dx_kernel<float>), dim3((unsigned)N), dim3(256),
+                     0, st, x, dy, gamma, mean, invstd, dx, D);
+}
+void launch_ln_bwd_dx_bf16(const rbf16* x, const rbf16* dy,
+                           const float* gamma, const float* mean,
+                           const float* invstd, rbf16* dx, long N, int D,
+                           hipStream_t st) {
+  hipLaunchKernelGGL((ln_bwd_dx_kernel<bf16>), dim3((unsigned)N), dim3(256),
+                     0, st, reinterpret_cast<const bf16*>(x),
+                     reinterpret_cast<const bf16*>(dy), gamma, mean, invstd,
+                     reinterpret_cast<bf16*>(dx), D);
+}
+
+void launch_ln_bwd_dgb_f32(const float* x, const float* dy, const float* mean,
+                           const float* invstd, float* dgamma, float* dbeta,
+                           long N, int D, hipStream_t st) {
+  int CT = ln_ct_for(D);
+  int RT = 256 / CT;
+  long rb = (N + RT - 1) / RT;
+  if (rb > 1024) rb = 1024;
+  dim3 grid((unsigned)(rb < 1 ? 1 : rb), (unsigned)((D + CT - 1) / CT));
+  hipLaunchKernelGGL((ln_bwd_dgb_kernel<float>), grid, dim3(256), 0, st, x,
+                     dy, mean, invstd, dgamma, dbeta, N, D, CT);
+}
+void launch_ln_bwd_dgb_bf16(const rbf16* x, const rbf16* dy,
+                            const float* mean, const float* invstd,
+                            float* dgamma, float* dbeta, long N, int D,
+                            hipStream_t st) {
+  int CT = ln_ct_for(D);
+  int RT = 256 / CT;
+  long rb = (N + RT - 1) / RT;
+  if (rb > 1024) rb = 1024;
+  dim3 grid((unsigned)(rb < 1 ? 1 : rb), (unsigned)((D + CT - 1) / CT));
+  hipLaunchKernelGGL((ln_bwd_dgb_kernel<bf16>), grid, dim3(256), 0, st,
+                     reinterpret_cast<const bf16*>(x),
+                     reinterpret_cast<const bf16*>(dy), mean, invstd, dgamma,
+                     dbeta, N, D, CT);
+}
