@@ -70,9 +70,13 @@ __device__ __forceinline__ void store8v<float>(float* p, const float* in) {
 // ---------------------------------------------------------------------------
 // stats: ws[0:C] += sum(x), ws[C:2C] += sum(x^2)
 // ---------------------------------------------------------------------------
+// Writes per-block partials to part[blockIdx.x * 2C + {c, C+c}] — plain
+// coalesced stores, NO global atomics (1024 serialized atomicAdds per
+// channel measured 130-300 GB/s on 7x7 layers). The finalize kernel
+// folds the <=kMaxRB partial rows.
 template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x,
-                                float* __restrict__ ws, long M, int C,
+                                float* __restrict__ part, long M, int C,
                                 int G) {
   __shared__ float lds[256 * 16];
   int cols = C >> 3;
@@ -84,7 +88,23 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
   if (cg < cols) {
     long rstride = (long)gridDim.x * RT;
     const T* base = x + cg * 8;
-    for (long r = (long)blockIdx.x * RT + rl; r < M; r += rstride) {
+    long r = (long)blockIdx.x * RT + rl;
+    for (; r + 3 * rstride < M; r += 4 * rstride) {  // 4 loads in flight
+      float v0[8], v1[8], v2[8], v3[8];
+      load8v<T>(base + r * C, v0);
+      load8v<T>(base + (r + rstride) * C, v1);
+      load8v<T>(base + (r + 2 * rstride) * C, v2);
+      load8v<T>(base + (r + 3 * rstride) * C, v3);
+#pragma unroll
+      for (int k = 0; k < 8; k++) {
+        s[k] += v0[k] + v1[k] + v2[k] + v3[k];
+        ss[k] = fmaf(v0[k], v0[k], ss[k]);
+        ss[k] = fmaf(v1[k], v1[k], ss[k]);
+        ss[k] = fmaf(v2[k], v2[k], ss[k]);
+        ss[k] = fmaf(v3[k], v3[k], ss[k]);
+      }
+    }
+    for (; r < M; r += rstride) {
       float v[8];
       load8v<T>(base + r * C, v);
 #pragma unroll
@@ -107,18 +127,19 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
     __syncthreads();
   }
   if (rl == 0 && cg < cols) {
+    float* slot = part + (long)blockIdx.x * 2 * C;
 #pragma unroll
     for (int k = 0; k < 8; k++) {
-      atomicAdd(&ws[cg * 8 + k], mine[k]);
-      atomicAdd(&ws[C + cg * 8 + k], mine[8 + k]);
+      slot[cg * 8 + k] = mine[k];
+      slot[C + cg * 8 + k] = mine[8 + k];
     }
   }
 }
 
-// scalar fallback (odd C)
+// scalar fallback (odd C): atomics into partial row 0 (zeroed by caller)
 template <typename T>
 __global__ void bn_stats_scalar_kernel(const T* __restrict__ x,
-                                       float* __restrict__ ws, long M,
+                                       float* __restrict__ part, long M,
                                        int C) {
   int c = blockIdx.y * blockDim.x + threadIdx.x;
   if (c >= C) return;
@@ -128,8 +149,8 @@ __global__ void bn_stats_scalar_kernel(const T* __restrict__ x,
     s += v;
     ss = fmaf(v, v, ss);
   }
-  atomicAdd(&ws[c], s);
-  atomicAdd(&ws[C + c], ss);
+  atomicAdd(&part[c], s);
+  atomicAdd(&part[C + c], ss);
 }
 
 // ---------------------------------------------------------------------------
@@ -159,7 +180,31 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     }
     long rstride = (long)gridDim.x * RT;
     long off = cg * 8;
-    for (long r = (long)blockIdx.x * RT + rl; r < M; r += rstride) {
+    long r = (long)blockIdx.x * RT + rl;
+    for (; r + rstride < M; r += 2 * rstride) {  // 2x(x,dy,y) in flight
+      float xv0[8], dv0[8], xv1[8], dv1[8];
+      load8v<T>(x + r * C + off, xv0);
+      load8v<T>(dy + r * C + off, dv0);
+      load8v<T>(x + (r + rstride) * C + off, xv1);
+      load8v<T>(dy + (r + rstride) * C + off, dv1);
+      if (RELU) {
+        float yv0[8], yv1[8];
+        load8v<T>(y + r * C + off, yv0);
+        load8v<T>(y + (r + rstride) * C + off, yv1);
+#pragma unroll
+        for (int k = 0; k < 8; k++) {
+          dv0[k] = yv0[k] > 0.f ? dv0[k] : 0.f;
+          dv1[k] = yv1[k] > 0.f ? dv1[k] : 0.f;
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < 8; k++) {
+        s1[k] += dv0[k] + dv1[k];
+        s2[k] = fmaf(dv0[k], (xv0[k] - mu[k]) * is[k], s2[k]);
+        s2[k] = fmaf(dv1[k], (xv1[k] - mu[k]) * is[k], s2[k]);
+      }
+    }
+    for (; r < M; r += rstride) {
       float xv[8], dv[8];
       load8v<T>(x + r * C + off, xv);
       load8v<T>(dy + r * C + off, dv);
@@ -189,10 +234,11 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     __syncthreads();
   }
   if (rl == 0 && cg < cols) {
+    float* slot = s12 + (long)blockIdx.x * 2 * C;
 #pragma unroll
     for (int k = 0; k < 8; k++) {
-      atomicAdd(&s12[cg * 8 + k], mine[k]);
-      atomicAdd(&s12[C + cg * 8 + k], mine[8 + k]);
+      slot[cg * 8 + k] = mine[k];
+      slot[C + cg * 8 + k] = mine[8 + k];
     }
   }
 }
@@ -240,7 +286,33 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
     sc[k] = scale[off + k];
     sh[k] = shift[off + k];
   }
-  for (long r = row0; r < M; r += rstride) {
+  long r = row0;
+  for (; r + rstride < M; r += 2 * rstride) {  // 2 rows in flight
+    float v0[8], v1[8];
+    load8v<T>(x + r * C + off, v0);
+    load8v<T>(x + (r + rstride) * C + off, v1);
+    float r0[8], r1[8];
+    if (ADD) {
+      load8v<T>(res + r * C + off, r0);
+      load8v<T>(res + (r + rstride) * C + off, r1);
+    }
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      v0[k] = fmaf(v0[k], sc[k], sh[k]);
+      v1[k] = fmaf(v1[k], sc[k], sh[k]);
+      if (ADD) {
+        v0[k] += r0[k];
+        v1[k] += r1[k];
+      }
+      if (RELU) {
+        v0[k] = fmaxf(v0[k], 0.f);
+        v1[k] = fmaxf(v1[k], 0.f);
+      }
+    }
+    store8v<T>(y + r * C + off, v0);
+    store8v<T>(y + (r + rstride) * C + off, v1);
+  }
+  for (; r < M; r += rstride) {
     float v[8];
     load8v<T>(x + r * C + off, v);
 #pragma unroll
@@ -342,15 +414,20 @@ __global__ void bn_bwd_apply_scalar_kernel(
 // ---------------------------------------------------------------------------
 template <typename P>
 __global__ void bn_finalize_train_kernel(
-    const float* __restrict__ ws,  // [2C]: sum, sumsq
-    const P* __restrict__ w, const P* __restrict__ b, P* __restrict__ rm,
-    P* __restrict__ rv, float momentum, float eps, long M,
+    const float* __restrict__ part,  // [rb][2C]: sum, sumsq partials
+    int rb, const P* __restrict__ w, const P* __restrict__ b,
+    P* __restrict__ rm, P* __restrict__ rv, float momentum, float eps, long M,
     float* __restrict__ out,  // [4C]: mean, invstd, scale, shift
     int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = ws[c] / M;
-  float var = fmaxf(ws[C + c] / M - mean * mean, 0.f);
+  float sum = 0.f, sumsq = 0.f;
+  for (int bk = 0; bk < rb; bk++) {
+    sum += part[(long)bk * 2 * C + c];
+    sumsq += part[(long)bk * 2 * C + C + c];
+  }
+  float mean = sum / M;
+  float var = fmaxf(sumsq / M - mean * mean, 0.f);
   float invstd = rsqrtf(var + eps);
   float scale = to_f32<P>(w[c]) * invstd;
   float shift = to_f32<P>(b[c]) - mean * scale;
@@ -382,14 +459,18 @@ __global__ void bn_finalize_eval_kernel(const P* __restrict__ w,
 
 template <typename P>
 __global__ void bn_bwd_finalize_kernel(
-    const float* __restrict__ s12,  // [2C]: sum_dy, sum_dy_xhat
-    const P* __restrict__ w, const float* __restrict__ mean,
+    const float* __restrict__ part,  // [rb][2C]: sum_dy, sum_dy_xhat
+    int rb, const P* __restrict__ w, const float* __restrict__ mean,
     const float* __restrict__ invstd, long M,
     float* __restrict__ abd,  // [3C]: A, B, D
     P* __restrict__ dgamma, P* __restrict__ dbeta, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float s1 = s12[c], s2 = s12[C + c];
+  float s1 = 0.f, s2 = 0.f;
+  for (int bk = 0; bk < rb; bk++) {
+    s1 += part[(long)bk * 2 * C + c];
+    s2 += part[(long)bk * 2 * C + C + c];
+  }
   float a = to_f32<P>(w[c]) * invstd[c];
   abd[c] = a;
   abd[C + c] = -a * (s2 / M) * invstd[c];
@@ -408,13 +489,23 @@ inline int g_for(int C) {  // col-groups per block: power of 2 (tree reduce)
   return g;
 }
 
-inline dim3 reduce_grid(long M, int C, int G) {
+// partial-row count: enough blocks for bandwidth, few enough that the
+// finalize reduce over rb stays cheap (exported via bn_reduce_rb below
+// so ext.cc sizes the partial buffer identically).
+inline int reduce_rb_impl(long M, int C) {
+  if (C % 8) return 1;  // scalar fallback accumulates into row 0
+  int G = g_for(C);
   int RT = 256 / G;
-  long rb = (M + RT - 1) / RT;
-  if (rb > 1024) rb = 1024;
+  long rb = (M + 8L * RT - 1) / (8L * RT);
+  if (rb > 320) rb = 320;
   if (rb < 1) rb = 1;
+  return (int)rb;
+}
+
+inline dim3 reduce_grid(long M, int C, int G) {
   int cols = C >> 3;
-  return dim3((unsigned)rb, (unsigned)((cols + G - 1) / G));
+  return dim3((unsigned)reduce_rb_impl(M, C),
+              (unsigned)((cols + G - 1) / G));
 }
 
 inline int apply_grid(long M, int C) {
@@ -426,6 +517,8 @@ inline int apply_grid(long M, int C) {
 }
 
 }  // namespace
+
+int bn_reduce_rb(long M, int C) { return reduce_rb_impl(M, C); }
 
 // ---- exported launchers ---------------------------------------------------
 
@@ -577,19 +670,19 @@ void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy, const rbf16* y,
 // ---- finalize launchers ---------------------------------------------------
 #define FINALIZE_GRID(C) dim3((unsigned)(((C) + 255) / 256)), dim3(256)
 
-void launch_bn_finalize_train_f32(const float* ws, const float* w,
+void launch_bn_finalize_train_f32(const float* part, int rb, const float* w,
                                   const float* b, float* rm, float* rv,
                                   float momentum, float eps, long M,
                                   float* out, int C, hipStream_t st) {
   hipLaunchKernelGGL((bn_finalize_train_kernel<float>), FINALIZE_GRID(C), 0,
-                     st, ws, w, b, rm, rv, momentum, eps, M, out, C);
+                     st, part, rb, w, b, rm, rv, momentum, eps, M, out, C);
 }
-void launch_bn_finalize_train_bf16(const float* ws, const rbf16* w,
+void launch_bn_finalize_train_bf16(const float* part, int rb, const rbf16* w,
                                    const rbf16* b, rbf16* rm, rbf16* rv,
                                    float momentum, float eps, long M,
                                    float* out, int C, hipStream_t st) {
   hipLaunchKernelGGL((bn_finalize_train_kernel<bf16>), FINALIZE_GRID(C), 0,
-                     st, ws, reinterpret_cast<const bf16*>(w),
+                     st, part, rb, reinterpret_cast<const bf16*>(w),
                      reinterpret_cast<const bf16*>(b),
                      reinterpret_cast<bf16*>(rm), reinterpret_cast<bf16*>(rv),
                      momentum, eps, M, out, C);
@@ -609,19 +702,19 @@ void launch_bn_finalize_eval_bf16(const rbf16* w, const rbf16* b,
                      reinterpret_cast<const bf16*>(rm),
                      reinterpret_cast<const bf16*>(rv), eps, out, C);
 }
-void launch_bn_bwd_finalize_f32(const float* s12, const float* w,
+void launch_bn_bwd_finalize_f32(const float* part, int rb, const float* w,
                                 const float* mean, const float* invstd,
                                 long M, float* abd, float* dgamma,
                                 float* dbeta, int C, hipStream_t st) {
   hipLaunchKernelGGL((bn_bwd_finalize_kernel<float>), FINALIZE_GRID(C), 0, st,
-                     s12, w, mean, invstd, M, abd, dgamma, dbeta, C);
+                     part, rb, w, mean, invstd, M, abd, dgamma, dbeta, C);
 }
-void launch_bn_bwd_finalize_bf16(const float* s12, const rbf16* w,
+void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
                                  const float* mean, const float* invstd,
                                  long M, float* abd, rbf16* dgamma,
                                  rbf16* dbeta, int C, hipStream_t st) {
   hipLaunchKernelGGL((bn_bwd_finalize_kernel<bf16>), FINALIZE_GRID(C), 0, st,
-                     s12, reinterpret_cast<const bf16*>(w), mean, invstd, M,
-                     abd, reinterpret_cast<bf16*>(dgamma),
+                     part, rb, reinterpret_cast<const bf16*>(w), mean, invstd,
+                     M, abd, reinterpret_cast<bf16*>(dgamma),
                      reinterpret_cast<bf16*>(dbeta), C);
 }

@@ -194,17 +194,20 @@ std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
   }
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
-  auto ws = at::zeros({2 * C}, opts);       // sum | sumsq
+  int rb = bn_reduce_rb(M, C);
+  auto ws = (C % 8 == 0) ? at::empty({(long)rb * 2 * C}, opts)
+                         : at::zeros({(long)rb * 2 * C}, opts);
   auto out = at::empty({4 * C}, opts);      // mean | invstd | scale | shift
   bn_stats(x, M, C, ws.data_ptr<float>());
   auto st = cur_stream(x);
   if (w.scalar_type() == at::kBFloat16)
-    launch_bn_finalize_train_bf16(ws.data_ptr<float>(), bf16_ptr(w),
+    launch_bn_finalize_train_bf16(ws.data_ptr<float>(), rb, bf16_ptr(w),
                                   bf16_ptr(b), bf16_ptr(rm), bf16_ptr(rv),
                                   (float)momentum, (float)eps, M,
                                   out.data_ptr<float>(), C, st);
   else
-    launch_bn_finalize_train_f32(ws.data_ptr<float>(), w.data_ptr<float>(),
+    launch_bn_finalize_train_f32(ws.data_ptr<float>(), rb,
+                                 w.data_ptr<float>(),
                                  b.data_ptr<float>(), rm.data_ptr<float>(),
                                  rv.data_ptr<float>(), (float)momentum,
                                  (float)eps, M, out.data_ptr<float>(), C, st);
@@ -246,7 +249,9 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   TORCH_CHECK(mean.is_contiguous() && invstd.is_contiguous());
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   auto opts = x.options().dtype(at::kFloat);
-  auto s12 = at::zeros({2 * C}, opts);
+  int rb = bn_reduce_rb(M, C);
+  auto s12 = (C % 8 == 0) ? at::empty({(long)rb * 2 * C}, opts)
+                          : at::zeros({(long)rb * 2 * C}, opts);
   auto abd = at::empty({3 * C}, opts);
   auto dgb = at::empty({2 * C}, x.options().dtype(w.scalar_type()));
   auto st = cur_stream(x);
@@ -261,14 +266,14 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                              invstd.data_ptr<float>(), s12.data_ptr<float>(),
                              M, C, relu, st);
   if (w.scalar_type() == at::kBFloat16)
-    launch_bn_bwd_finalize_bf16(s12.data_ptr<float>(), bf16_ptr(w),
+    launch_bn_bwd_finalize_bf16(s12.data_ptr<float>(), rb, bf16_ptr(w),
                                 mean.data_ptr<float>(),
                                 invstd.data_ptr<float>(), M,
                                 abd.data_ptr<float>(), bf16_ptr(dgb),
                                 bf16_ptr(dgb) + C, C, st);
   else
-    launch_bn_bwd_finalize_f32(s12.data_ptr<float>(), w.data_ptr<float>(),
-                               mean.data_ptr<float>(),
+    launch_bn_bwd_finalize_f32(s12.data_ptr<float>(), rb,
+                               w.data_ptr<float>(), mean.data_ptr<float>(),
                                invstd.data_ptr<float>(), M,
                                abd.data_ptr<float>(), dgb.data_ptr<float>(),
                                dgb.data_ptr<float>() + C, C, st);

@@ -82,11 +82,12 @@ void launch_ln_bwd_dgb_bf16(const rbf16* x, const rbf16* dy,
                             hipStream_t st);
 
 // bn_relu.hip finalize kernels
-void launch_bn_finalize_train_f32(const float* ws, const float* w,
+int bn_reduce_rb(long M, int C);  // partial rows the reduce kernels emit
+void launch_bn_finalize_train_f32(const float* part, int rb, const float* w,
                                   const float* b, float* rm, float* rv,
                                   float momentum, float eps, long M,
                                   float* out, int C, hipStream_t st);
-void launch_bn_finalize_train_bf16(const float* ws, const rbf16* w,
+void launch_bn_finalize_train_bf16(const float* part, int rb, const rbf16* w,
                                    const rbf16* b, rbf16* rm, rbf16* rv,
                                    float momentum, float eps, long M,
                                    float* out, int C, hipStream_t st);
@@ -96,11 +97,11 @@ void launch_bn_finalize_eval_f32(const float* w, const float* b,
 void launch_bn_finalize_eval_bf16(const rbf16* w, const rbf16* b,
                                   const rbf16* rm, const rbf16* rv, float eps,
                                   float* out, int C, hipStream_t st);
-void launch_bn_bwd_finalize_f32(const float* s12, const float* w,
+void launch_bn_bwd_finalize_f32(const float* part, int rb, const float* w,
                                 const float* mean, const float* invstd,
                                 long M, float* abd, float* dgamma,
                                 float* dbeta, int C, hipStream_t st);
-void launch_bn_bwd_finalize_bf16(const float* s12, const rbf16* w,
+void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
                                  const float* mean, const float* invstd,
                                  long M, float* abd, rbf16* dgamma,
                                  rbf16* dbeta, int C, hipStream_t st);
