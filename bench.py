@@ -42,6 +42,9 @@ def parse_args():
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--torch-profile", type=str, default=None,
                     help="write a torch.profiler summary of the timed steps")
+    ap.add_argument("--hip-graph", action="store_true",
+                    help="capture the train step in a hipGraph and replay "
+                         "(sync SGD/momentum only; removes launch gaps)")
     return ap.parse_args()
 
 
@@ -118,7 +121,7 @@ def main():
     n_batches = 4
     batches = [make_batch(1000 * comm.rank + i) for i in range(n_batches)]
 
-    def step(i):
+    def eager_step(i):
         x, y = batches[i % n_batches]
         engine.zero_grad()
         if hasattr(opt, "begin_step"):
@@ -126,6 +129,38 @@ def main():
         loss = model.loss(x, y)
         engine.step(loss, opt)
         return loss
+
+    step = eager_step
+    if args.hip_graph and use_cuda:
+        # capture fwd+bwd+comm+fused-optimizer in one hipGraph; replay
+        # per step with a device-side copy of the rotating batch.
+        x_s = batches[0][0].clone()
+        y_s = batches[0][1].clone()
+
+        def body():
+            engine.zero_grad()
+            loss = model.loss(x_s, y_s)
+            engine.step(loss, opt)
+            return loss
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                loss_static = body()
+        torch.cuda.current_stream().wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            loss_static = body()
+
+        def graph_step(i):
+            x, y = batches[i % n_batches]
+            x_s.copy_(x)
+            y_s.copy_(y)
+            graph.replay()
+            return loss_static
+
+        step = graph_step
 
     for i in range(args.warmup):
         step(i)

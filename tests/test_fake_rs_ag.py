@@ -1,0 +1,140 @@
+"""rs_ag engine path (reduce-scatter + shard optimizer + all-gather —
+the RCCL recast of the reference's PS sync mode, C4) validated on CPU
+through the in-process fake collective backend."""
+
+import threading
+
+import pytest
+import torch
+
+from rocps.engine.fake_comm import FakeComm, FakeWorld, run_fake_cluster
+from rocps.engine.sync_engine import SyncEngine
+from rocps.models import mnist_mlp, synthetic_mnist_batch
+from rocps.train.optimizer import GradientDescentOptimizer, MomentumOptimizer
+
+
+def _ref_run(steps, global_batch, lr, momentum=None, seed=3):
+    torch.manual_seed(7)
+    m = mnist_mlp(hidden=32)
+    opt = (torch.optim.SGD(m.parameters(), lr=lr, momentum=momentum)
+           if momentum else torch.optim.SGD(m.parameters(), lr=lr))
+    losses = []
+    for s in range(steps):
+        g = torch.Generator().manual_seed(seed + s)
+        x, y = synthetic_mnist_batch(global_batch, generator=g)
+        opt.zero_grad()
+        loss = m.loss(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.detach()))
+    return m, losses
+
+
+def _fake_run(world, steps, per_rank_batch, lr, momentum=None, mode="rs_ag",
+              seed=3):
+    results = {}
+    lock = threading.Lock()
+    # build models on the MAIN thread: torch.manual_seed is process-global,
+    # seeding from rank threads races the RNG draws of model init
+    models = []
+    for _ in range(world):
+        torch.manual_seed(7)
+        models.append(mnist_mlp(hidden=32))
+
+    def rank_fn(comm: FakeComm, rank: int):
+        m = models[rank]
+        eng = SyncEngine(m, comm, mode=mode, bucket_cap_mb=0.01)
+        opt = (MomentumOptimizer(lr, momentum) if momentum
+               else GradientDescentOptimizer(lr))
+        eng.broadcast_parameters(src=0)
+        losses = []
+        for s in range(steps):
+            g = torch.Generator().manual_seed(seed + s)
+            x, y = synthetic_mnist_batch(per_rank_batch * world,
+                                         generator=g)
+            lo = rank * per_rank_batch
+            xb, yb = x[lo:lo + per_rank_batch], y[lo:lo + per_rank_batch]
+            eng.zero_grad()
+            loss = m.loss(xb, yb)
+            eng.step(loss, opt)
+            losses.append(float(loss.detach()))
+        with lock:
+            results[rank] = (m, losses)
+        return rank
+
+    run_fake_cluster(world, rank_fn)
+    return results
+
+
+@pytest.mark.parametrize("mode", ["rs_ag", "allreduce"])
+def test_fake_two_ranks_match_single_process(mode):
+    steps, B, lr = 8, 16, 0.2
+    ref_model, _ = _ref_run(steps, 2 * B, lr)
+    results = _fake_run(2, steps, B, lr, mode=mode)
+    p_ref = dict(ref_model.named_parameters())
+    for rank in (0, 1):
+        m, _ = results[rank]
+        for n, p in m.named_parameters():
+            assert torch.allclose(p, p_ref[n], atol=1e-5), (mode, rank, n)
+    # both ranks end with identical replicas
+    m0, m1 = results[0][0], results[1][0]
+    for (n, a), (_, b) in zip(m0.named_parameters(), m1.named_parameters()):
+        assert torch.equal(a, b), n
+
+
+def test_fake_four_ranks_momentum_rs_ag():
+    steps, B, lr, mu = 6, 8, 0.1, 0.9
+    ref_model, _ = _ref_run(steps, 4 * B, lr, momentum=mu)
+    results = _fake_run(4, steps, B, lr, momentum=mu, mode="rs_ag")
+    p_ref = dict(ref_model.named_parameters())
+    m, _ = results[0]
+    for n, p in m.named_parameters():
+        assert torch.allclose(p, p_ref[n], atol=1e-5), n
+
+
+def test_fake_collectives_primitives():
+    def rank_fn(comm: FakeComm, rank: int):
+        t = torch.full((4,), float(rank + 1))
+        comm.all_reduce(t)
+        assert torch.equal(t, torch.full((4,), 6.0))  # 1+2+3
+        inp = torch.arange(6, dtype=torch.float32) + 10 * rank
+        out = torch.empty(2)
+        comm.reduce_scatter(out, inp)
+        expect = (torch.arange(6, dtype=torch.float32) * 3 + 30)[
+            rank * 2:rank * 2 + 2]
+        assert torch.equal(out, expect)
+        full = torch.empty(6)
+        comm.all_gather(full, out)
+        b = torch.full((2,), float(rank))
+        comm.broadcast(b, src=1)
+        assert torch.equal(b, torch.full((2,), 1.0))
+        return True
+
+    assert run_fake_cluster(3, rank_fn) == [True, True, True]
+
+
+def test_fake_fault_injection_surfaces_error():
+    def rank_fn(comm_args):
+        pass
+
+    world = FakeWorld(2, timeout=5.0)
+    errors = []
+
+    def runner(rank):
+        comm = FakeComm(world, rank, fail_at=3 if rank == 1 else None)
+        try:
+            for _ in range(5):
+                t = torch.ones(2)
+                comm.all_reduce(t)
+        except BaseException as e:  # noqa: BLE001
+            errors.append((rank, type(e).__name__))
+            world.barrier.abort()
+
+    ts = [threading.Thread(target=runner, args=(r,)) for r in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(20)
+    kinds = {k for _, k in errors}
+    assert "ConnectionError" in kinds  # the injected failure
+    assert len(errors) == 2  # the peer observes a broken collective
