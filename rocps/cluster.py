@@ -264,6 +264,12 @@ class Server:
     The master task additionally hosts cluster-wide coordination state.
     Extra services (async-PS shard push/pull) register op handlers via
     :meth:`register_handler`.
+
+    The server itself is native (csrc/control/control_plane.cc — the C1
+    [NATIVE] obligation): coordination ops run in C++ threads with no
+    GIL; unknown ops call back into the registered Python handlers.
+    A pure-Python server (``ROCPS_PURE_PY_SERVER=1`` or missing
+    extension) keeps the same protocol for reference and tests.
     """
 
     def __init__(self, cluster: Union[ClusterSpec, Dict[str, Any]],
@@ -281,28 +287,76 @@ class Server:
         self._handlers: Dict[str, Callable] = {}
         self._shutdown_evt = threading.Event()
         self._tcp: Optional[_TCPServer] = None
+        self._native = None
         self._thread: Optional[threading.Thread] = None
         if start:
             self.start()
 
     # -- lifecycle ----------------------------------------------------------
     def start(self) -> None:
-        if self._tcp is not None:
+        if self._tcp is not None or self._native is not None:
             return
         bind_host = os.environ.get("ROCPS_BIND_HOST", "0.0.0.0")
-        self._tcp = _TCPServer((bind_host, self._port), _Handler)
-        if self._port == 0:  # ephemeral: record actual port
-            self._port = self._tcp.server_address[1]
-        self._tcp.rocps_server = self  # type: ignore[attr-defined]
-        self._thread = threading.Thread(
-            target=self._tcp.serve_forever, name="rocps-server", daemon=True)
-        self._thread.start()
+        native = None
+        if not os.environ.get("ROCPS_PURE_PY_SERVER"):
+            try:
+                from rocps import _ccp
+                native = _ccp.ControlPlaneServer(self._port, self.is_master,
+                                 self.job_name, self.task_index)
+            except ImportError:
+                native = None
+        if native is not None:
+            native.set_fallback(self._native_fallback)
+            self._port = native.start("" if bind_host == "0.0.0.0"
+                                      else bind_host)
+            self._native = native
+            self._watch = threading.Thread(target=self._watch_native,
+                                           daemon=True,
+                                           name="rocps-native-watch")
+            self._watch.start()
+            # stop native threads before interpreter finalization: a C++
+            # thread re-acquiring the GIL after Py_Finalize aborts
+            import atexit
+            atexit.register(self.stop)
+        else:
+            self._tcp = _TCPServer((bind_host, self._port), _Handler)
+            if self._port == 0:  # ephemeral: record actual port
+                self._port = self._tcp.server_address[1]
+            self._tcp.rocps_server = self  # type: ignore[attr-defined]
+            self._thread = threading.Thread(
+                target=self._tcp.serve_forever, name="rocps-server",
+                daemon=True)
+            self._thread.start()
         # Register with the cluster master (self-registration included).
         try:
             ControlClient(self.cluster.master_address).register(
                 self.job_name, self.task_index)
         except Exception:
             pass  # master may not be up yet; heartbeats will register later
+
+    def _watch_native(self) -> None:
+        self._native.wait_shutdown(0.0)
+        self._shutdown_evt.set()
+
+    def _native_fallback(self, head_bytes: bytes, payload: bytes):
+        """Called from a C++ connection thread for non-builtin ops."""
+        try:
+            head = json.loads(head_bytes)
+        except Exception:
+            return (json.dumps({"ok": False, "error": "bad json"}).encode(),
+                    b"")
+        op = head.get("op", "")
+        fn = self._handlers.get(op)
+        if fn is None:
+            return (json.dumps(
+                {"ok": False,
+                 "error": f"not master; no handler for {op!r}"}).encode(),
+                b"")
+        try:
+            reply, rpayload = fn(head, payload)
+            return json.dumps(reply).encode(), bytes(rpayload or b"")
+        except Exception as e:  # surface handler errors to the caller
+            return (json.dumps({"ok": False, "error": repr(e)}).encode(), b"")
 
     @property
     def target(self) -> str:
@@ -316,8 +370,14 @@ class Server:
         """Block until the cluster is shut down (the reference's ps loop)."""
         self._shutdown_evt.wait(timeout)
 
+    def shutdown_requested(self) -> bool:
+        return self._shutdown_evt.is_set()
+
     def stop(self) -> None:
         self._shutdown_evt.set()
+        if self._native is not None:
+            self._native.stop()
+            self._native = None
         if self._tcp is not None:
             self._tcp.shutdown()
             self._tcp.server_close()

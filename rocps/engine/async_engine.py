@@ -192,7 +192,7 @@ class AsyncPSShard:
         W = self.num_workers
         with self.coord.stop_on_exception():
             while not (self.coord.should_stop()
-                       or self.server._shutdown_evt.is_set()):
+                       or self.server.shutdown_requested()):
                 progressed = False
                 for w in range(W):
                     seq, ack = int(h[8 + 2 * w]), int(h[8 + 2 * w + 1])
@@ -306,11 +306,19 @@ class AsyncWorkerEngine:
                 view = (flat.view(torch.uint16) if self.dtype == torch.bfloat16
                         else flat)
                 payload = view.numpy().tobytes()
-                r = c.call("init_shard", payload=payload,
-                           numel=flat.numel(),
-                           dtype=str(self.dtype).replace("torch.", ""),
-                           optimizer=optimizer_spec(optimizer))
-                if not r.get("ok"):
+                deadline = time.time() + 60.0
+                while True:
+                    r = c.call("init_shard", payload=payload,
+                               numel=flat.numel(),
+                               dtype=str(self.dtype).replace("torch.", ""),
+                               optimizer=optimizer_spec(optimizer))
+                    if r.get("ok"):
+                        break
+                    # PS may still be wiring its shard service (startup race)
+                    if ("no handler" in str(r.get("error", ""))
+                            and time.time() < deadline):
+                        time.sleep(0.2)
+                        continue
                     raise RuntimeError(f"init_shard failed: {r}")
                 c.close()
         self._attach()

@@ -29,6 +29,11 @@ ext_modules = [
         sources=["csrc/bundle/bundle.cc"],
         extra_compile_args=["-O2", "-std=c++17"],
     ),
+    CppExtension(
+        name="rocps._ccp",
+        sources=["csrc/control/control_plane.cc"],
+        extra_compile_args=["-O2", "-std=c++17"],
+    ),
 ]
 
 _hip_sources = [
