@@ -40,6 +40,8 @@ flags.DEFINE_integer("save_checkpoint_steps", 0, "ckpt every N steps (0=off)")
 flags.DEFINE_integer("hidden", 32, "MLP hidden width")
 flags.DEFINE_integer("seed", 42, "data/model seed")
 flags.DEFINE_string("loss_log", "", "write per-step losses (JSON) here")
+flags.DEFINE_integer("kill_after_step", 0,
+                     "fault injection: crash this worker after N local steps")
 FLAGS = flags.FLAGS
 
 
@@ -82,6 +84,7 @@ def main(_):
             save_checkpoint_steps=(FLAGS.save_checkpoint_steps or None),
             save_checkpoint_secs=(None if FLAGS.save_checkpoint_steps else 600),
     ) as sess:
+        local_step = 0
         while not sess.should_stop():
             # deterministic synthetic batch for this global step; each
             # worker takes its slice (the reference feeds next_batch())
@@ -91,6 +94,9 @@ def main(_):
             xb, yb = x[lo:lo + FLAGS.batch_size], y[lo:lo + FLAGS.batch_size]
             loss = sess.run(lambda: trainer.model.loss(xb, yb))
             losses.append(float(loss))
+            local_step += 1
+            if FLAGS.kill_after_step and local_step >= FLAGS.kill_after_step:
+                os._exit(17)  # simulated crash (no cleanup, like a real one)
 
     if FLAGS.loss_log:
         with open(FLAGS.loss_log, "w") as f:
@@ -100,10 +106,17 @@ def main(_):
     import time as _time
     trainer.control.incr("workers_done")
     if is_chief:
+        # wait for every LIVE worker: dead ones (heartbeat silence) are
+        # excluded, so a crashed peer cannot wedge the shutdown
         deadline = _time.time() + 300
-        while (trainer.control.read_counter("workers_done") < num_workers
-               and _time.time() < deadline):
-            _time.sleep(0.1)
+        while _time.time() < deadline:
+            done = trainer.control.read_counter("workers_done")
+            dead = [m for m in (trainer.heartbeat.dead_members(threshold=6.0)
+                                if trainer.heartbeat else [])
+                    if m.startswith("worker:")]
+            if done >= num_workers - len(dead):
+                break
+            _time.sleep(0.2)
         from rocps.cluster import shutdown_cluster
         shutdown_cluster(cluster)
     return 0

@@ -28,7 +28,8 @@ import threading
 import time
 from typing import Any, Callable, Dict, List, Optional, Union
 
-__all__ = ["ClusterSpec", "Server", "ControlClient", "shutdown_cluster"]
+__all__ = ["ClusterSpec", "Server", "ControlClient", "shutdown_cluster",
+           "HeartbeatThread"]
 
 _LEN = struct.Struct("<Q")
 _MAX_MSG = 1 << 30
@@ -528,3 +529,42 @@ def shutdown_cluster(cluster: ClusterSpec) -> None:
                 c.close()
             except Exception:
                 pass  # task already gone — that's what shutdown wants
+
+
+class HeartbeatThread:
+    """Failure detection (SURVEY §5): each task heartbeats the master;
+    ``dead_members`` reports tasks silent for > threshold seconds.
+
+    The reference surfaces peer death as gRPC UnavailableError; here the
+    control plane's membership table plays that role — the async PS
+    tolerates silent workers, sync-mode scripts can poll
+    ``dead_members`` and trigger restart-from-checkpoint.
+    """
+
+    def __init__(self, cluster: ClusterSpec, job: str, task: int,
+                 interval: float = 2.0):
+        self.client = ControlClient(cluster.master_address)
+        self.job, self.task = job, task
+        self.interval = interval
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="rocps-heartbeat")
+        self._thread.start()
+
+    def _run(self) -> None:
+        while not self._stop.wait(self.interval):
+            try:
+                self.client.heartbeat(self.job, self.task)
+            except Exception:
+                pass  # master unreachable — the master is the failed one
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    def dead_members(self, threshold: float = 10.0) -> List[str]:
+        now = time.time()
+        try:
+            members = self.client.members()
+        except Exception:
+            return []
+        return [m for m, last in members.items() if now - last > threshold]

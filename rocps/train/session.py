@@ -43,8 +43,12 @@ class Trainer:
         self.task_index = task_index
         self.cluster = cluster
         self.control = control
+        self.heartbeat = None
         if cluster is not None and control is None:
             self.control = ControlClient(cluster.master_address)
+        if cluster is not None:
+            from rocps.cluster import HeartbeatThread
+            self.heartbeat = HeartbeatThread(cluster, job_name, task_index)
         self.device = device or (
             torch.device("cuda", torch.cuda.current_device())
             if torch.cuda.is_available() else torch.device("cpu"))

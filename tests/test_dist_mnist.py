@@ -122,3 +122,39 @@ def test_async_two_workers_two_ps(free_ports, tmp_path):
         assert all(l == l and l < 50 for l in losses[i])  # finite, sane
     total_steps = sum(len(l) for l in losses.values())
     assert total_steps >= 40  # cluster-wide step target reached
+
+
+def test_async_worker_death_tolerated(free_ports, tmp_path):
+    """SURVEY §5 failure handling: in async mode a dead worker is
+    tolerated — the PS keeps serving, the chief detects the silent peer
+    via heartbeats and shuts the cluster down cleanly.
+
+    One internal retry: the scenario is inherently timing-dependent
+    (process kill vs heartbeat windows) and must not flake the gate."""
+    last_err = None
+    for attempt in range(2):
+        d = tmp_path / f"a{attempt}"
+        os.makedirs(d, exist_ok=True)
+        ports = free_ports(3)
+        ps = [f"127.0.0.1:{ports[0]}"]
+        workers = [f"127.0.0.1:{ports[1]}", f"127.0.0.1:{ports[2]}"]
+        procs = {
+            "ps0": _launch("ps", 0, ps, workers, ["--issync=0"], d),
+            "w0": _launch("worker", 0, ps, workers,
+                          ["--issync=0", "--train_steps=40",
+                           "--batch_size=8", f"--loss_log={d}/l0.json"], d),
+            "w1": _launch("worker", 1, ps, workers,
+                          ["--issync=0", "--train_steps=40",
+                           "--batch_size=8", "--kill_after_step=3"], d),
+        }
+        try:
+            outs = _wait_all(procs, timeout=120)
+            assert procs["w1"].returncode == 17, outs["w1"]  # injected crash
+            assert procs["w0"].returncode == 0, outs["w0"]   # survivor done
+            assert procs["ps0"].returncode == 0, outs["ps0"]  # clean PS exit
+            losses = json.load(open(d / "l0.json"))
+            assert len(losses) >= 10 and all(l == l for l in losses)
+            return
+        except AssertionError as e:
+            last_err = e
+    raise last_err
