@@ -331,9 +331,28 @@ class AsyncWorkerEngine:
             itemsize = torch.empty(0, dtype=dtype).element_size()
             layout = _ShmLayout(int(info["numel"]),
                                 self.cluster.num_tasks("worker"), itemsize)
-            self.segs[s] = _ShmSeg(info["path"], layout, dtype, create=False)
+            seg = _ShmSeg(info["path"], layout, dtype, create=False)
+            if self.device.type == "cuda":
+                self._pin_segment(seg, layout)
+            self.segs[s] = seg
         self._attached = True
         self.pull()
+
+    @staticmethod
+    def _pin_segment(seg: _ShmSeg, layout: _ShmLayout) -> None:
+        """hipHostRegister the mapped shard so push/pull run as true
+        pinned DMA on the copy engine (hipMemcpyAsync on a
+        side stream). Falls back to pageable copies if registration is
+        refused."""
+        try:
+            import ctypes
+            base = ctypes.addressof(ctypes.c_char.from_buffer(seg.mm))
+            r = torch.cuda.cudart().cudaHostRegister(base, layout.total, 0)
+            if int(r) != 0:
+                raise RuntimeError(f"cudaHostRegister rc={int(r)}")
+            seg.pinned = True
+        except Exception:
+            seg.pinned = False
 
     # -- engine interface (used by Trainer) ------------------------------
     def broadcast_parameters(self, src: int = 0) -> None:
@@ -352,11 +371,10 @@ class AsyncWorkerEngine:
     def push_and_pull(self) -> None:
         self.local_step += 1
         w = self.task_index
-        # push grads to every shard, bump seq
+        # push grads to every shard, bump seq (D2H straight into the
+        # shard slot — pinned DMA when _pin_segment succeeded)
         for s, seg in enumerate(self.segs):
-            seg.grads[w].copy_(self.flat_grads[s].cpu()
-                               if self.device.type != "cpu"
-                               else self.flat_grads[s])
+            seg.grads[w].copy_(self.flat_grads[s])
             seg.header[8 + 2 * w] = self.local_step
         # wait for own applies (the reference's RunGraph round trip)
         deadline = time.time() + 120.0
@@ -371,9 +389,7 @@ class AsyncWorkerEngine:
 
     def pull(self) -> None:
         for s, seg in enumerate(self.segs):
-            self.flat_params[s].copy_(seg.params.to(self.device)
-                                      if self.device.type != "cpu"
-                                      else seg.params)
+            self.flat_params[s].copy_(seg.params)
 
     def detach(self) -> None:
         for seg in self.segs:
