@@ -91,7 +91,8 @@ def main(_):
             g = torch.Generator().manual_seed(FLAGS.seed + trainer.global_step)
             x, y = synthetic_mnist_batch(global_batch, generator=g)
             lo = FLAGS.task_index * FLAGS.batch_size
-            xb, yb = x[lo:lo + FLAGS.batch_size], y[lo:lo + FLAGS.batch_size]
+            xb = x[lo:lo + FLAGS.batch_size].to(trainer.device)
+            yb = y[lo:lo + FLAGS.batch_size].to(trainer.device)
             loss = sess.run(lambda: trainer.model.loss(xb, yb))
             losses.append(float(loss))
             local_step += 1

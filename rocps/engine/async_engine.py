@@ -258,13 +258,15 @@ class AsyncWorkerEngine:
     def __init__(self, model: torch.nn.Module, cluster: ClusterSpec,
                  control: ControlClient, task_index: int,
                  device: torch.device, num_shards: Optional[int] = None,
-                 strategy: str = "greedy"):
+                 strategy: str = "greedy",
+                 staleness_bound: Optional[int] = None):
         self.model = model
         self.cluster = cluster
         self.control = control
         self.task_index = task_index
         self.device = device
         self.num_shards = num_shards or cluster.num_tasks("ps")
+        self.staleness_bound = staleness_bound
         named = [(n, p) for n, p in model.named_parameters()
                  if p.requires_grad]
         self.plan = plan_shards(named, self.num_shards, strategy=strategy)
@@ -311,7 +313,8 @@ class AsyncWorkerEngine:
                     r = c.call("init_shard", payload=payload,
                                numel=flat.numel(),
                                dtype=str(self.dtype).replace("torch.", ""),
-                               optimizer=optimizer_spec(optimizer))
+                               optimizer=optimizer_spec(optimizer),
+                               staleness_bound=self.staleness_bound)
                     if r.get("ok"):
                         break
                     # PS may still be wiring its shard service (startup race)

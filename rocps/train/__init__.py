@@ -20,3 +20,4 @@ from rocps.train.saver import Saver, latest_checkpoint  # noqa: F401
 from rocps.train.coordinator import Coordinator, LooseVersionThread  # noqa: F401
 from rocps.train.global_step import (  # noqa: F401
     get_or_create_global_step, get_global_step)
+from rocps.train.supervisor import Supervisor  # noqa: F401

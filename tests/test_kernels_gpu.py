@@ -310,3 +310,40 @@ def test_bn_add_relu_fused(dtype):
     assert torch.allclose(rg.grad.float().cpu(), rc.grad.float(), atol=tol,
                           rtol=tol)
     assert torch.allclose(bg.grad.cpu(), bc.grad, atol=5e-3, rtol=1e-3)
+
+
+def test_kernels_deterministic():
+    """SURVEY §5 race discipline: with atomic-free reductions the rocps
+    kernels are bitwise deterministic across identical runs."""
+    from rocps.ops.norm import fused_bn_act
+    from rocps.ops.softmax_xent import softmax_cross_entropy_with_logits
+    from rocps.ops.layer_norm import layer_norm
+    torch.manual_seed(0)
+    x = (torch.randn(16, 64, 28, 28, device=DEV).bfloat16()
+         .to(memory_format=torch.channels_last))
+    w = torch.rand(64, device=DEV) + 0.5
+    b = torch.randn(64, device=DEV)
+    outs = []
+    for _ in range(2):
+        xg = x.clone().requires_grad_(True)
+        y = fused_bn_act(xg, w.clone().requires_grad_(True),
+                         b.clone().requires_grad_(True),
+                         torch.zeros(64, device=DEV),
+                         torch.ones(64, device=DEV), training=True,
+                         relu=True)
+        y.sum().backward()
+        outs.append((y.detach().clone(), xg.grad.detach().clone()))
+    torch.cuda.synchronize()
+    assert torch.equal(outs[0][0], outs[1][0])
+    assert torch.equal(outs[0][1], outs[1][1])
+
+    lg = torch.randn(64, 1000, device=DEV).bfloat16()
+    lab = torch.randint(0, 1000, (64,), device=DEV)
+    l1 = softmax_cross_entropy_with_logits(lg, lab)
+    l2 = softmax_cross_entropy_with_logits(lg, lab)
+    assert torch.equal(l1, l2)
+
+    xl = torch.randn(128, 768, device=DEV).bfloat16()
+    wl = torch.rand(768, device=DEV)
+    bl = torch.randn(768, device=DEV)
+    assert torch.equal(layer_norm(xl, wl, bl), layer_norm(xl, wl, bl))

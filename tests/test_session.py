@@ -102,3 +102,45 @@ def test_step_counter_hook():
     sc = StepCounterHook(every_n_steps=2, batch_size=16)
     _run(tr, 6, hooks=[sc])
     assert sc.last_steps_per_sec is not None and sc.last_steps_per_sec > 0
+
+
+def test_supervisor_shim(tmp_path):
+    import rocps
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    sv = rocps.train.Supervisor(is_chief=True, logdir=str(tmp_path / "sv"),
+                                trainer=tr, save_model_secs=0)
+    g = torch.Generator().manual_seed(0)
+    with sv.managed_session() as sess:
+        for _ in range(3):
+            x, y = synthetic_mnist_batch(8, generator=g)
+            sess.run(lambda: tr.model.loss(x, y))
+    assert tr.global_step == 3
+
+
+def test_event_file_writer(tmp_path):
+    import glob
+    import struct
+
+    import numpy as np
+
+    from rocps import _bundle as C
+    from rocps.obs import EventFileWriter
+    w = EventFileWriter(str(tmp_path / "ev"))
+    for s in range(5):
+        w.add_scalar("loss", 2.0 - 0.1 * s, s)
+    w.close()
+    files = glob.glob(str(tmp_path / "ev" / "events.out.tfevents.*"))
+    assert len(files) == 1
+    raw = open(files[0], "rb").read()
+    off, n = 0, 0
+    while off < len(raw):
+        (ln,) = struct.unpack("<Q", raw[off:off + 8])
+        (lc,) = struct.unpack("<I", raw[off + 8:off + 12])
+        assert lc == C.masked_crc32c(
+            np.frombuffer(raw[off:off + 8], dtype=np.uint8))
+        data = raw[off + 12:off + 12 + ln]
+        (dc,) = struct.unpack("<I", raw[off + 12 + ln:off + 16 + ln])
+        assert dc == C.masked_crc32c(np.frombuffer(data, dtype=np.uint8))
+        off += 16 + ln
+        n += 1
+    assert n == 6  # file_version + 5 scalars
