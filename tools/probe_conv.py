@@ -36,3 +36,19 @@ for fmt_name, fmt in [("NHWC", torch.channels_last), ("NCHW", torch.contiguous_f
         us = timeit(fb)
         flops = 2 * N * Co * Ci * k * k * (H // s) * (W // s) * 3  # fwd+dgrad+wgrad
         print(f"{fmt_name} {N}x{Ci}x{H}x{W} -> {Co} k{k}s{s}: {us:8.0f} us  {flops/us/1e6:7.1f} TF/s", flush=True)
+
+# stem variants: Cin=3 is pathological for implicit GEMM — probe padding
+print("--- stem Cin pad variants (NHWC) ---", flush=True)
+for Ci in (3, 4, 8):
+    conv = torch.nn.Conv2d(Ci, 64, 7, stride=2, padding=3, bias=False)
+    conv = conv.to(dev).bfloat16().to(memory_format=torch.channels_last)
+    x = torch.randn(256, Ci, 224, 224, device=dev).bfloat16().to(
+        memory_format=torch.channels_last).requires_grad_(True)
+    def fb():
+        y = conv(x)
+        y.backward(torch.ones_like(y))
+        x.grad = None
+        conv.weight.grad = None
+    us = timeit(fb)
+    flops = 2 * 256 * 64 * 3 * 49 * 112 * 112 * 3
+    print(f"stem Ci={Ci}: {us:8.0f} us  ({flops/us/1e6:6.1f} TF/s vs Ci=3 work)", flush=True)
