@@ -412,42 +412,28 @@ __global__ void bn_bwd_apply_scalar_kernel(
 // ---------------------------------------------------------------------------
 // per-channel finalize kernels (C threads, one launch)
 // ---------------------------------------------------------------------------
-// thread layout (CT c-lanes x J row-lanes, CT pow2): the rb-partial fold
-// is parallel over J with an LDS tree — a serial rb loop on C threads
-// measured 73 us/call (one block for C=64).
+// wave-per-channel rb-fold: 64 lanes sweep the partial rows of ONE
+// channel + shuffle-reduce. The CTxJ-with-J=4 layout before this still
+// measured 64 us/call (C=64 -> one block, 80 serial loads per lane).
 template <typename P>
 __global__ void bn_finalize_train_kernel(
     const float* __restrict__ part,  // [rb][2C]: sum, sumsq partials
     int rb, const P* __restrict__ w, const P* __restrict__ b,
     P* __restrict__ rm, P* __restrict__ rv, float momentum, float eps, long M,
     float* __restrict__ out,  // [4C]: mean, invstd, scale, shift
-    int C, int CT) {
-  __shared__ float lsum[256];
-  __shared__ float lss[256];
-  int J = blockDim.x / CT;
-  int cl = threadIdx.x % CT;
-  int j = threadIdx.x / CT;
-  int c = blockIdx.x * CT + cl;
+    int C) {
+  int wave = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int c = blockIdx.x * (blockDim.x / WAVE) + wave;
+  if (c >= C) return;
   float sum = 0.f, sumsq = 0.f;
-  if (c < C) {
-    for (int bk = j; bk < rb; bk += J) {
-      sum += part[(long)bk * 2 * C + c];
-      sumsq += part[(long)bk * 2 * C + C + c];
-    }
+  for (int bk = lane; bk < rb; bk += WAVE) {
+    sum += part[(long)bk * 2 * C + c];
+    sumsq += part[(long)bk * 2 * C + C + c];
   }
-  lsum[threadIdx.x] = sum;
-  lss[threadIdx.x] = sumsq;
-  __syncthreads();
-  for (int step = (J >> 1) * CT; step >= CT; step >>= 1) {
-    if ((int)threadIdx.x < step) {
-      lsum[threadIdx.x] += lsum[threadIdx.x + step];
-      lss[threadIdx.x] += lss[threadIdx.x + step];
-    }
-    __syncthreads();
-  }
-  if (j != 0 || c >= C) return;
-  sum = lsum[cl];
-  sumsq = lss[cl];
+  sum = wave_reduce_sum(sum);
+  sumsq = wave_reduce_sum(sumsq);
+  if (lane != 0) return;
   float mean = sum / M;
   float var = fmaxf(sumsq / M - mean * mean, 0.f);
   float invstd = rsqrtf(var + eps);
@@ -485,33 +471,19 @@ __global__ void bn_bwd_finalize_kernel(
     int rb, const P* __restrict__ w, const float* __restrict__ mean,
     const float* __restrict__ invstd, long M,
     float* __restrict__ abd,  // [3C]: A, B, D
-    P* __restrict__ dgamma, P* __restrict__ dbeta, int C, int CT) {
-  __shared__ float l1[256];
-  __shared__ float l2[256];
-  int J = blockDim.x / CT;
-  int cl = threadIdx.x % CT;
-  int j = threadIdx.x / CT;
-  int c = blockIdx.x * CT + cl;
+    P* __restrict__ dgamma, P* __restrict__ dbeta, int C) {
+  int wave = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  int c = blockIdx.x * (blockDim.x / WAVE) + wave;
+  if (c >= C) return;
   float s1 = 0.f, s2 = 0.f;
-  if (c < C) {
-    for (int bk = j; bk < rb; bk += J) {
-      s1 += part[(long)bk * 2 * C + c];
-      s2 += part[(long)bk * 2 * C + C + c];
-    }
+  for (int bk = lane; bk < rb; bk += WAVE) {
+    s1 += part[(long)bk * 2 * C + c];
+    s2 += part[(long)bk * 2 * C + C + c];
   }
-  l1[threadIdx.x] = s1;
-  l2[threadIdx.x] = s2;
-  __syncthreads();
-  for (int step = (J >> 1) * CT; step >= CT; step >>= 1) {
-    if ((int)threadIdx.x < step) {
-      l1[threadIdx.x] += l1[threadIdx.x + step];
-      l2[threadIdx.x] += l2[threadIdx.x + step];
-    }
-    __syncthreads();
-  }
-  if (j != 0 || c >= C) return;
-  s1 = l1[cl];
-  s2 = l2[cl];
+  s1 = wave_reduce_sum(s1);
+  s2 = wave_reduce_sum(s2);
+  if (lane != 0) return;
   float a = to_f32<P>(w[c]) * invstd[c];
   abd[c] = a;
   abd[C + c] = -a * (s2 / M) * invstd[c];
@@ -709,32 +681,26 @@ void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy, const rbf16* y,
 }
 
 // ---- finalize launchers ---------------------------------------------------
-static int fin_ct(int C) {  // pow2 c-lanes per 256-thread block
-  int ct = 8;
-  while (ct < C && ct < 256) ct <<= 1;
-  return ct;
-}
-#define FINALIZE_GRID(C, CT) dim3((unsigned)(((C) + (CT) - 1) / (CT))), dim3(256)
+#define FIN_WPB 4  // waves (=channels) per 256-thread block
+#define FINALIZE_GRID(C, CT) \
+  dim3((unsigned)(((C) + FIN_WPB - 1) / FIN_WPB)), dim3(256)
 
 void launch_bn_finalize_train_f32(const float* part, int rb, const float* w,
                                   const float* b, float* rm, float* rv,
                                   float momentum, float eps, long M,
                                   float* out, int C, hipStream_t st) {
-  int CT = fin_ct(C);
-  hipLaunchKernelGGL((bn_finalize_train_kernel<float>), FINALIZE_GRID(C, CT),
-                     0, st, part, rb, w, b, rm, rv, momentum, eps, M, out, C,
-                     CT);
+  hipLaunchKernelGGL((bn_finalize_train_kernel<float>), FINALIZE_GRID(C, 0),
+                     0, st, part, rb, w, b, rm, rv, momentum, eps, M, out, C);
 }
 void launch_bn_finalize_train_bf16(const float* part, int rb, const rbf16* w,
                                    const rbf16* b, rbf16* rm, rbf16* rv,
                                    float momentum, float eps, long M,
                                    float* out, int C, hipStream_t st) {
-  int CT = fin_ct(C);
-  hipLaunchKernelGGL((bn_finalize_train_kernel<bf16>), FINALIZE_GRID(C, CT),
+  hipLaunchKernelGGL((bn_finalize_train_kernel<bf16>), FINALIZE_GRID(C, 0),
                      0, st, part, rb, reinterpret_cast<const bf16*>(w),
                      reinterpret_cast<const bf16*>(b),
                      reinterpret_cast<bf16*>(rm), reinterpret_cast<bf16*>(rv),
-                     momentum, eps, M, out, C, CT);
+                     momentum, eps, M, out, C);
 }
 void launch_bn_finalize_eval_f32(const float* w, const float* b,
                                  const float* rm, const float* rv, float eps,
@@ -755,18 +721,15 @@ void launch_bn_bwd_finalize_f32(const float* part, int rb, const float* w,
                                 const float* mean, const float* invstd,
                                 long M, float* abd, float* dgamma,
                                 float* dbeta, int C, hipStream_t st) {
-  int CT = fin_ct(C);
-  hipLaunchKernelGGL((bn_bwd_finalize_kernel<float>), FINALIZE_GRID(C, CT), 0,
-                     st, part, rb, w, mean, invstd, M, abd, dgamma, dbeta, C,
-                     CT);
+  hipLaunchKernelGGL((bn_bwd_finalize_kernel<float>), FINALIZE_GRID(C, 0), 0,
+                     st, part, rb, w, mean, invstd, M, abd, dgamma, dbeta, C);
 }
 void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
                                  const float* mean, const float* invstd,
                                  long M, float* abd, rbf16* dgamma,
                                  rbf16* dbeta, int C, hipStream_t st) {
-  int CT = fin_ct(C);
-  hipLaunchKernelGGL((bn_bwd_finalize_kernel<bf16>), FINALIZE_GRID(C, CT), 0,
+  hipLaunchKernelGGL((bn_bwd_finalize_kernel<bf16>), FINALIZE_GRID(C, 0), 0,
                      st, part, rb, reinterpret_cast<const bf16*>(w), mean,
                      invstd, M, abd, reinterpret_cast<bf16*>(dgamma),
-                     reinterpret_cast<bf16*>(dbeta), C, CT);
+                     reinterpret_cast<bf16*>(dbeta), C);
 }
