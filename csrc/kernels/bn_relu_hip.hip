@@ -156,12 +156,17 @@ __global__ void bn_stats_scalar_kernel(const T* __restrict__ x,
 // ---------------------------------------------------------------------------
 // bwd reduce: s12[0:C] += sum(dy_m), s12[C:2C] += sum(dy_m * xhat)
 // ---------------------------------------------------------------------------
-template <typename T, bool RELU>
+// RELU && !USE_Y: the mask is recomputed as fmaf(x, scale, shift) > 0 —
+// bitwise the forward's pre-activation — so the y tensor is never read
+// (saves one full pass in reduce and one in apply for plain-ReLU BNs).
+template <typename T, bool RELU, bool USE_Y>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ dy,
                                      const T* __restrict__ y,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
+                                     const float* __restrict__ scale,
+                                     const float* __restrict__ shift,
                                      float* __restrict__ s12, long M, int C,
                                      int G) {
   __shared__ float lds[256 * 16];
@@ -172,11 +177,15 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
   int cg = blockIdx.y * G + gl;
   float s1[8] = {0.f}, s2[8] = {0.f};
   if (cg < cols) {
-    float mu[8], is[8];
+    float mu[8], is[8], sc[8], sh[8];
 #pragma unroll
     for (int k = 0; k < 8; k++) {
       mu[k] = mean[cg * 8 + k];
       is[k] = invstd[cg * 8 + k];
+      if (RELU && !USE_Y) {
+        sc[k] = scale[cg * 8 + k];
+        sh[k] = shift[cg * 8 + k];
+      }
     }
     long rstride = (long)gridDim.x * RT;
     long off = cg * 8;
@@ -187,7 +196,7 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
       load8v<T>(dy + r * C + off, dv0);
       load8v<T>(x + (r + rstride) * C + off, xv1);
       load8v<T>(dy + (r + rstride) * C + off, dv1);
-      if (RELU) {
+      if (RELU && USE_Y) {
         float yv0[8], yv1[8];
         load8v<T>(y + r * C + off, yv0);
         load8v<T>(y + (r + rstride) * C + off, yv1);
@@ -195,6 +204,12 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
         for (int k = 0; k < 8; k++) {
           dv0[k] = yv0[k] > 0.f ? dv0[k] : 0.f;
           dv1[k] = yv1[k] > 0.f ? dv1[k] : 0.f;
+        }
+      } else if (RELU) {
+#pragma unroll
+        for (int k = 0; k < 8; k++) {
+          dv0[k] = fmaf(xv0[k], sc[k], sh[k]) > 0.f ? dv0[k] : 0.f;
+          dv1[k] = fmaf(xv1[k], sc[k], sh[k]) > 0.f ? dv1[k] : 0.f;
         }
       }
 #pragma unroll
@@ -208,11 +223,15 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
       float xv[8], dv[8];
       load8v<T>(x + r * C + off, xv);
       load8v<T>(dy + r * C + off, dv);
-      if (RELU) {
+      if (RELU && USE_Y) {
         float yv[8];
         load8v<T>(y + r * C + off, yv);
 #pragma unroll
         for (int k = 0; k < 8; k++) dv[k] = yv[k] > 0.f ? dv[k] : 0.f;
+      } else if (RELU) {
+#pragma unroll
+        for (int k = 0; k < 8; k++)
+          dv[k] = fmaf(xv[k], sc[k], sh[k]) > 0.f ? dv[k] : 0.f;
       }
 #pragma unroll
       for (int k = 0; k < 8; k++) {
@@ -351,12 +370,14 @@ __global__ void bn_apply_scalar_kernel(const T* __restrict__ x,
 // ---------------------------------------------------------------------------
 // bwd apply: dx = A*dy_m + B*x + D; (ADD) dres = dy_m
 // ---------------------------------------------------------------------------
-template <typename T, bool RELU, bool ADD>
+template <typename T, bool RELU, bool ADD, bool USE_Y>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const T* __restrict__ dy,
                                     const T* __restrict__ y,
                                     T* __restrict__ dx, T* __restrict__ dres,
-                                    const float* __restrict__ abd, long M,
+                                    const float* __restrict__ abd,
+                                    const float* __restrict__ scale,
+                                    const float* __restrict__ shift, long M,
                                     int C) {
   int cols = C >> 3;
   long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -366,22 +387,30 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
   int cg = (int)(tid % cols);
   long row0 = tid / cols;
   long off = (long)cg * 8;
-  float A[8], B[8], D[8];
+  float A[8], B[8], D[8], sc[8], sh[8];
 #pragma unroll
   for (int k = 0; k < 8; k++) {
     A[k] = abd[off + k];
     B[k] = abd[C + off + k];
     D[k] = abd[2 * C + off + k];
+    if (RELU && !USE_Y) {
+      sc[k] = scale[off + k];
+      sh[k] = shift[off + k];
+    }
   }
   for (long r = row0; r < M; r += rstride) {
     float xv[8], dv[8];
     load8v<T>(x + r * C + off, xv);
     load8v<T>(dy + r * C + off, dv);
-    if (RELU) {
+    if (RELU && USE_Y) {
       float yv[8];
       load8v<T>(y + r * C + off, yv);
 #pragma unroll
       for (int k = 0; k < 8; k++) dv[k] = yv[k] > 0.f ? dv[k] : 0.f;
+    } else if (RELU) {
+#pragma unroll
+      for (int k = 0; k < 8; k++)
+        dv[k] = fmaf(xv[k], sc[k], sh[k]) > 0.f ? dv[k] : 0.f;
     }
     if (ADD) store8v<T>(dres + r * C + off, dv);
     float o[8];
@@ -557,17 +586,21 @@ void launch_bn_stats_bf16(const rbf16* x, float* ws, long M, int C,
 
 template <typename T>
 static void _bwd_reduce(const T* x, const T* dy, const T* y,
-                        const float* mean, const float* invstd, float* s12,
-                        long M, int C, bool relu, hipStream_t stream) {
+                        const float* mean, const float* invstd,
+                        const float* scale, const float* shift, float* s12,
+                        long M, int C, bool relu, bool use_y,
+                        hipStream_t stream) {
   if (C % 8 == 0) {
     int G = g_for(C);
     dim3 grid = reduce_grid(M, C, G);
-    if (relu)
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid, dim3(256), 0,
-                         stream, x, dy, y, mean, invstd, s12, M, C, G);
-    else
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid, dim3(256), 0,
-                         stream, x, dy, y, mean, invstd, s12, M, C, G);
+#define BRK(R, U)                                                           \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, R, U>), grid, dim3(256), 0, \
+                       stream, x, dy, y, mean, invstd, scale, shift, s12,   \
+                       M, C, G)
+    if (relu && use_y) BRK(true, true);
+    else if (relu) BRK(true, false);
+    else BRK(false, true);
+#undef BRK
   } else {
     dim3 grid((unsigned)(M < 512 ? (M ? M : 1) : 512),
               (unsigned)((C + 255) / 256));
@@ -583,18 +616,21 @@ static void _bwd_reduce(const T* x, const T* dy, const T* y,
 }
 void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
                               const float* mean, const float* invstd,
+                              const float* scale, const float* shift,
                               float* s12, long M, int C, bool relu,
-                              hipStream_t st) {
-  _bwd_reduce(x, dy, y, mean, invstd, s12, M, C, relu, st);
+                              bool use_y, hipStream_t st) {
+  _bwd_reduce(x, dy, y, mean, invstd, scale, shift, s12, M, C, relu, use_y,
+              st);
 }
 void launch_bn_bwd_reduce_bf16(const rbf16* x, const rbf16* dy,
                                const rbf16* y, const float* mean,
-                               const float* invstd, float* s12, long M, int C,
-                               bool relu, hipStream_t st) {
+                               const float* invstd, const float* scale,
+                               const float* shift, float* s12, long M, int C,
+                               bool relu, bool use_y, hipStream_t st) {
   _bwd_reduce(reinterpret_cast<const bf16*>(x),
               reinterpret_cast<const bf16*>(dy),
-              reinterpret_cast<const bf16*>(y), mean, invstd, s12, M, C, relu,
-              st);
+              reinterpret_cast<const bf16*>(y), mean, invstd, scale, shift,
+              s12, M, C, relu, use_y, st);
 }
 
 template <typename T>
@@ -640,18 +676,21 @@ void launch_bn_apply_bf16(const rbf16* x, const rbf16* res, rbf16* y,
 
 template <typename T>
 static void _bwd_apply(const T* x, const T* dy, const T* y, T* dx, T* dres,
-                       const float* abd, long M, int C, bool relu,
-                       hipStream_t stream) {
+                       const float* abd, const float* scale,
+                       const float* shift, long M, int C, bool relu,
+                       bool use_y, hipStream_t stream) {
   bool add = dres != nullptr;
   if (C % 8 == 0) {
     int grid = apply_grid(M, C);
-#define BAP(R, A)                                                            \
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, R, A>), dim3(grid),           \
-                       dim3(256), 0, stream, x, dy, y, dx, dres, abd, M, C)
-    if (relu && add) BAP(true, true);
-    else if (relu) BAP(true, false);
-    else if (add) BAP(false, true);
-    else BAP(false, false);
+#define BAP(R, A, U)                                                         \
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<T, R, A, U>), dim3(grid),        \
+                       dim3(256), 0, stream, x, dy, y, dx, dres, abd, scale, \
+                       shift, M, C)
+    if (relu && add) BAP(true, true, true);
+    else if (relu && use_y) BAP(true, false, true);
+    else if (relu) BAP(true, false, false);
+    else if (add) BAP(false, true, true);
+    else BAP(false, false, true);
 #undef BAP
   } else {
     int grid = (int)((M * C + 255) / 256 < 2048 ? ((M * C + 255) / 256 ?: 1)
@@ -667,17 +706,20 @@ static void _bwd_apply(const T* x, const T* dy, const T* y, T* dx, T* dres,
   }
 }
 void launch_bn_bwd_apply_f32(const float* x, const float* dy, const float* y,
-                             float* dx, float* dres, const float* abd, long M,
-                             int C, bool relu, hipStream_t st) {
-  _bwd_apply(x, dy, y, dx, dres, abd, M, C, relu, st);
+                             float* dx, float* dres, const float* abd,
+                             const float* scale, const float* shift, long M,
+                             int C, bool relu, bool use_y, hipStream_t st) {
+  _bwd_apply(x, dy, y, dx, dres, abd, scale, shift, M, C, relu, use_y, st);
 }
 void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy, const rbf16* y,
                               rbf16* dx, rbf16* dres, const float* abd,
-                              long M, int C, bool relu, hipStream_t st) {
+                              const float* scale, const float* shift, long M,
+                              int C, bool relu, bool use_y, hipStream_t st) {
   _bwd_apply(reinterpret_cast<const bf16*>(x),
              reinterpret_cast<const bf16*>(dy),
              reinterpret_cast<const bf16*>(y), reinterpret_cast<bf16*>(dx),
-             reinterpret_cast<bf16*>(dres), abd, M, C, relu, st);
+             reinterpret_cast<bf16*>(dres), abd, scale, shift, M, C, relu,
+             use_y, st);
 }
 
 // ---- finalize launchers ---------------------------------------------------

@@ -213,7 +213,9 @@ std::vector<at::Tensor> bn_relu_fwd_train(at::Tensor x, at::Tensor w,
                                  (float)eps, M, out.data_ptr<float>(), C, st);
   auto y = bn_apply(x, res ? &*res : nullptr, out.data_ptr<float>() + 2 * C,
                     out.data_ptr<float>() + 3 * C, M, C, relu);
-  return {y, out.narrow(0, 0, C), out.narrow(0, C, C)};
+  // 4th return: the whole [4C] fold tensor — backward can recompute the
+  // ReLU mask from x*scale+shift and skip reading y entirely.
+  return {y, out.narrow(0, 0, C), out.narrow(0, C, C), out};
 }
 
 at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
@@ -241,7 +243,18 @@ at::Tensor bn_relu_fwd_eval(at::Tensor x, at::Tensor w, at::Tensor b,
 std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                                     at::Tensor w, at::Tensor mean,
                                     at::Tensor invstd, bool relu,
-                                    bool has_res) {
+                                    bool has_res,
+                                    std::optional<at::Tensor> fold4c) {
+  // fold4c = the forward's [4C] {mean|invstd|scale|shift} tensor; with it
+  // (and no fused residual) the ReLU mask recomputes from x — y unread.
+  bool use_y = !(relu && !has_res && fold4c.has_value());
+  const float* sc = nullptr;
+  const float* sh = nullptr;
+  if (!use_y) {
+    int Cf = (int)x.size(1);
+    sc = fold4c->data_ptr<float>() + 2 * Cf;
+    sh = fold4c->data_ptr<float>() + 3 * Cf;
+  }
   long M;
   int C;
   bn_shapes(x, &M, &C);
@@ -258,13 +271,13 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
   if (x.scalar_type() == at::kBFloat16)
     launch_bn_bwd_reduce_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
                               mean.data_ptr<float>(),
-                              invstd.data_ptr<float>(),
-                              s12.data_ptr<float>(), M, C, relu, st);
+                              invstd.data_ptr<float>(), sc, sh,
+                              s12.data_ptr<float>(), M, C, relu, use_y, st);
   else
     launch_bn_bwd_reduce_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
                              y.data_ptr<float>(), mean.data_ptr<float>(),
-                             invstd.data_ptr<float>(), s12.data_ptr<float>(),
-                             M, C, relu, st);
+                             invstd.data_ptr<float>(), sc, sh,
+                             s12.data_ptr<float>(), M, C, relu, use_y, st);
   if (w.scalar_type() == at::kBFloat16)
     launch_bn_bwd_finalize_bf16(s12.data_ptr<float>(), rb, bf16_ptr(w),
                                 mean.data_ptr<float>(),
@@ -284,12 +297,14 @@ std::vector<at::Tensor> bn_relu_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
     launch_bn_bwd_apply_bf16(bf16_ptr(x), bf16_ptr(dy), bf16_ptr(y),
                              bf16_ptr(dx),
                              has_res ? bf16_ptr(dres) : nullptr,
-                             abd.data_ptr<float>(), M, C, relu, st);
+                             abd.data_ptr<float>(), sc, sh, M, C, relu,
+                             use_y, st);
   else
     launch_bn_bwd_apply_f32(x.data_ptr<float>(), dy.data_ptr<float>(),
                             y.data_ptr<float>(), dx.data_ptr<float>(),
                             has_res ? dres.data_ptr<float>() : nullptr,
-                            abd.data_ptr<float>(), M, C, relu, st);
+                            abd.data_ptr<float>(), sc, sh, M, C, relu,
+                            use_y, st);
   if (has_res)
     return {dx, dgb.narrow(0, 0, C), dgb.narrow(0, C, C), dres};
   return {dx, dgb.narrow(0, 0, C), dgb.narrow(0, C, C)};

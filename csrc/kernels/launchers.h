@@ -46,18 +46,22 @@ void launch_bn_apply_bf16(const rbf16* x, const rbf16* res, rbf16* y,
                           bool relu, hipStream_t st);
 void launch_bn_bwd_reduce_f32(const float* x, const float* dy, const float* y,
                               const float* mean, const float* invstd,
+                              const float* scale, const float* shift,
                               float* s12, long M, int C, bool relu,
-                              hipStream_t st);
+                              bool use_y, hipStream_t st);
 void launch_bn_bwd_reduce_bf16(const rbf16* x, const rbf16* dy,
                                const rbf16* y, const float* mean,
-                               const float* invstd, float* s12, long M, int C,
-                               bool relu, hipStream_t st);
+                               const float* invstd, const float* scale,
+                               const float* shift, float* s12, long M, int C,
+                               bool relu, bool use_y, hipStream_t st);
 void launch_bn_bwd_apply_f32(const float* x, const float* dy, const float* y,
-                             float* dx, float* dres, const float* abd, long M,
-                             int C, bool relu, hipStream_t st);
+                             float* dx, float* dres, const float* abd,
+                             const float* scale, const float* shift, long M,
+                             int C, bool relu, bool use_y, hipStream_t st);
 void launch_bn_bwd_apply_bf16(const rbf16* x, const rbf16* dy, const rbf16* y,
                               rbf16* dx, rbf16* dres, const float* abd,
-                              long M, int C, bool relu, hipStream_t st);
+                              const float* scale, const float* shift, long M,
+                              int C, bool relu, bool use_y, hipStream_t st);
 
 // layernorm.hip
 void launch_ln_fwd_f32(const float* x, const float* gamma, const float* beta,

@@ -41,16 +41,19 @@ class _FusedBNAct(torch.autograd.Function):
             res = (residual.contiguous(memory_format=torch.channels_last)
                    if has_res else None)
             if training:
-                y, save_mean, save_invstd = hip_ext().bn_relu_fwd_train(
-                    xc, weight, bias, running_mean, running_var,
-                    momentum, eps, relu, res)
+                y, save_mean, save_invstd, fold4c = \
+                    hip_ext().bn_relu_fwd_train(
+                        xc, weight, bias, running_mean, running_var,
+                        momentum, eps, relu, res)
             else:
                 y = hip_ext().bn_relu_fwd_eval(
                     xc, weight, bias, running_mean, running_var, eps, relu,
                     res)
                 save_mean = running_mean.float()
                 save_invstd = (running_var.float() + eps).rsqrt()
-            ctx.save_for_backward(xc, weight, save_mean, save_invstd, y)
+                fold4c = None
+            ctx.save_for_backward(xc, weight, save_mean, save_invstd, y,
+                                  *( [fold4c] if fold4c is not None else []))
             ctx.hip = True
             return y
         # CPU fp32 reference path
@@ -83,11 +86,13 @@ class _FusedBNAct(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, mean, invstd, y = ctx.saved_tensors
+        x, weight, mean, invstd, y = ctx.saved_tensors[:5]
+        fold4c = ctx.saved_tensors[5] if len(ctx.saved_tensors) > 5 else None
         if ctx.hip:
             dyc = dy.contiguous(memory_format=torch.channels_last)
             outs = hip_ext().bn_relu_bwd(
-                x, dyc, y, weight, mean, invstd, ctx.relu, ctx.has_res)
+                x, dyc, y, weight, mean, invstd, ctx.relu, ctx.has_res,
+                fold4c)
             dx, dgamma, dbeta = outs[0], outs[1], outs[2]
             dres = outs[3] if ctx.has_res else None
             return (dx, dres, dgamma, dbeta,
