@@ -372,6 +372,36 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
 }
 
 // ---------------------------------------------------------------------------
+// MFMA bf16 GEMM (C10): D = A @ W^T (+bias) (+gelu)
+// ---------------------------------------------------------------------------
+
+at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
+                     std::optional<at::Tensor> bias, long epilogue) {
+  TORCH_CHECK(A.is_cuda() && A.dim() == 2 && A.is_contiguous());
+  TORCH_CHECK(W.is_cuda() && W.dim() == 2 && W.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16
+              && W.scalar_type() == at::kBFloat16);
+  long M = A.size(0), K = A.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
+              "gemm_bf16 needs M%128==0, N%128==0, K%64==0 (got ", M, "x",
+              N, "x", K, ")");
+  const float* bptr = nullptr;
+  at::Tensor b32;
+  if (epilogue >= 1) {
+    TORCH_CHECK(bias.has_value(), "bias epilogue needs a bias tensor");
+    b32 = bias->to(at::kFloat).contiguous();
+    TORCH_CHECK(b32.numel() == N);
+    bptr = b32.data_ptr<float>();
+  }
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(A.device());
+  auto D = at::empty({M, N}, A.options());
+  launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), (int)M,
+                   (int)N, (int)K, (int)epilogue, cur_stream(A));
+  return D;
+}
+
+// ---------------------------------------------------------------------------
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "rocps hand-written CDNA4 (gfx950) kernels";
@@ -384,4 +414,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_relu_bwd", &bn_relu_bwd);
   m.def("ln_fwd", &ln_fwd);
   m.def("ln_bwd", &ln_bwd);
+  m.def("gemm_bf16", &gemm_bf16);
 }

@@ -109,3 +109,8 @@ void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
                                  const float* mean, const float* invstd,
                                  long M, float* abd, rbf16* dgamma,
                                  rbf16* dbeta, int C, hipStream_t st);
+
+// gemm_bf16.hip
+void launch_gemm_bf16(const rbf16* A, const rbf16* W, const float* bias,
+                      rbf16* D, int M, int N, int K, int epilogue,
+                      hipStream_t st);

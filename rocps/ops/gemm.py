@@ -1,0 +1,64 @@
+"""Hand-written MFMA bf16 GEMM — capability C10.
+
+``matmul_nt(a, w)`` computes a @ w.T (the torch Linear contract) on the
+CDNA4 MFMA kernel (csrc/kernels/gemm_bf16.hip: 128x128 macro-tile,
+global_load_lds staging, XOR-swizzled LDS, 16x16x32 bf16 MFMA), with
+optional fused bias and tanh-GELU epilogues — the SURVEY C10 "bias+GELU
+epilogue variant for BERT".
+
+Plain (unfused) GEMMs in the models run through rocBLAS/hipBLASLt via
+torch.nn.Linear, per the MI355X design rules; this kernel is the fused
+forward building block and the shape contract is M%128 == N%128 == 0,
+K%64 == 0 (use ``supported(...)`` to check).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from rocps.ops import hip_ext
+
+__all__ = ["matmul_nt", "linear_gelu", "supported"]
+
+
+def supported(m: int, n: int, k: int) -> bool:
+    return m % 128 == 0 and n % 128 == 0 and k % 64 == 0
+
+
+def matmul_nt(a: torch.Tensor, w: torch.Tensor,
+              bias: Optional[torch.Tensor] = None,
+              gelu: bool = False) -> torch.Tensor:
+    """a (M,K) bf16 @ w (N,K) bf16 -> (M,N) bf16, fp32 accumulation."""
+    epi = 2 if gelu else (1 if bias is not None else 0)
+    return hip_ext().gemm_bf16(a, w, bias, epi)
+
+
+class _LinearGeluFn(torch.autograd.Function):
+    """Fused fwd (MFMA GEMM + bias + GELU); backward recomputes the
+    pre-activation with a library GEMM (plain GEMMs may use rocBLAS)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w, b)
+        return matmul_nt(x, w, b, gelu=True)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, b = ctx.saved_tensors
+        pre = (x @ w.t()).float() + b.float()
+        c = 0.7978845608028654
+        t = torch.tanh(c * (pre + 0.044715 * pre ** 3))
+        dgelu = 0.5 * (1 + t) + 0.5 * pre * (1 - t * t) * c * (
+            1 + 3 * 0.044715 * pre ** 2)
+        g = (dy.float() * dgelu).to(dy.dtype)
+        dx = g @ w
+        dw = g.t() @ x
+        db = g.sum(0)
+        return dx, dw, db.to(b.dtype)
+
+
+def linear_gelu(x: torch.Tensor, w: torch.Tensor,
+                b: torch.Tensor) -> torch.Tensor:
+    return _LinearGeluFn.apply(x, w, b)

@@ -347,3 +347,60 @@ def test_kernels_deterministic():
     wl = torch.rand(768, device=DEV)
     bl = torch.randn(768, device=DEV)
     assert torch.equal(layer_norm(xl, wl, bl), layer_norm(xl, wl, bl))
+
+
+# ---------------------------------------------------------------------------
+# MFMA bf16 GEMM (C10)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("shape", [(128, 128, 64), (256, 384, 128),
+                                   (1024, 768, 768), (4096, 3072, 768)])
+def test_gemm_bf16_matmul_nt(shape):
+    """vs torch matmul (rocBLAS) — asymmetric operands (guide G9: a
+    symmetric B passes a transposed C-write)."""
+    from rocps.ops.gemm import matmul_nt
+    M, N, K = shape
+    torch.manual_seed(1)
+    a = (torch.randn(M, K, device=DEV) / K ** 0.25).bfloat16()
+    w = (torch.randn(N, K, device=DEV) / K ** 0.25).bfloat16()
+    got = matmul_nt(a, w)
+    torch.cuda.synchronize()
+    ref = (a.float() @ w.float().t())
+    diff = (got.float() - ref).abs()
+    denom = ref.abs().clamp_min(1.0)
+    assert (diff / denom).max().item() < 2e-2, (diff / denom).max().item()
+
+
+def test_gemm_bf16_bias_gelu_epilogue():
+    from rocps.ops.gemm import matmul_nt
+    M, N, K = 256, 256, 128
+    torch.manual_seed(2)
+    a = (torch.randn(M, K, device=DEV) / 4).bfloat16()
+    w = (torch.randn(N, K, device=DEV) / 4).bfloat16()
+    b = torch.randn(N, device=DEV)
+    got = matmul_nt(a, w, bias=b, gelu=True)
+    torch.cuda.synchronize()
+    pre = a.float() @ w.float().t() + b
+    ref = torch.nn.functional.gelu(pre, approximate="tanh")
+    assert torch.allclose(got.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_linear_gelu_autograd():
+    from rocps.ops.gemm import linear_gelu
+    M, N, K = 128, 128, 64
+    torch.manual_seed(3)
+    x = (torch.randn(M, K, device=DEV) / 4).bfloat16().requires_grad_(True)
+    w = (torch.randn(N, K, device=DEV) / 4).bfloat16().requires_grad_(True)
+    b = torch.randn(N, device=DEV).bfloat16().requires_grad_(True)
+    y = linear_gelu(x, w, b)
+    g = torch.randn_like(y)
+    y.backward(g)
+    # fp32 reference
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    b2 = b.detach().float().requires_grad_(True)
+    y2 = torch.nn.functional.gelu(x2 @ w2.t() + b2, approximate="tanh")
+    y2.backward(g.float())
+    assert torch.allclose(y.float(), y2, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(x.grad.float(), x2.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(w.grad.float(), w2.grad, atol=5e-2, rtol=5e-2)

@@ -110,7 +110,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--reps", type=int, default=50)
     ap.add_argument("--what", default="all",
-                    choices=["all", "bn", "optim", "softmax", "ln"])
+                    choices=["all", "bn", "optim", "softmax", "ln", "gemm"])
     args = ap.parse_args()
     dev = torch.device("cuda:0")
     torch.manual_seed(0)
@@ -122,7 +122,25 @@ def main():
         bench_softmax(dev, args.reps)
     if args.what in ("all", "ln"):
         bench_ln(dev, args.reps)
+    if args.what in ("all", "gemm"):
+        bench_gemm(dev, args.reps)
 
 
 if __name__ == "__main__":
     main()
+
+
+def bench_gemm(dev, reps):
+    from rocps.ops.gemm import matmul_nt
+    for (M, N, K) in [(4096, 4096, 4096), (8192, 768, 768),
+                      (8192, 3072, 768), (8192, 768, 3072)]:
+        a = (torch.randn(M, K, device=dev) / 8).bfloat16()
+        w = (torch.randn(N, K, device=dev) / 8).bfloat16()
+        us = timeit(lambda: matmul_nt(a, w), reps)
+        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
+        print(f"rocps gemm_bf16 {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
+              flush=True)
+        us = timeit(lambda: a @ w.t(), reps)
+        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
+        print(f"rocblas matmul  {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
+              flush=True)
