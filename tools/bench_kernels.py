@@ -53,10 +53,10 @@ def bench_bn(dev, dtype, reps):
         us = timeit(lambda: ext.bn_relu_fwd_train(x, w, b, rm, rv, 0.1, 1e-5,
                                                   True, None), reps)
         line(f"bn_fwd_train(stats+fin+apply) {tag}", us, 3 * nb)
-        y, mean, invstd = ext.bn_relu_fwd_train(x, w, b, rm, rv, 0.1, 1e-5,
-                                                True, None)
+        y, mean, invstd, fold4c = ext.bn_relu_fwd_train(
+            x, w, b, rm, rv, 0.1, 1e-5, True, None)
         us = timeit(lambda: ext.bn_relu_bwd(x, dy, y, w, mean, invstd, True,
-                                            False), reps)
+                                            False, fold4c), reps)
         line(f"bn_bwd(reduce+fin+apply)      {tag}", us, 7 * nb)
 
         # torch-native composition for the same op
@@ -106,6 +106,22 @@ def bench_ln(dev, reps):
             line(f"ln_fwd {N}x{D} {str(dtype)[6:]}", us, 2 * N * D * es)
 
 
+def bench_gemm(dev, reps):
+    from rocps.ops.gemm import matmul_nt
+    for (M, N, K) in [(4096, 4096, 4096), (8192, 768, 768),
+                      (8192, 3072, 768), (8192, 768, 3072)]:
+        a = (torch.randn(M, K, device=dev) / 8).bfloat16()
+        w = (torch.randn(N, K, device=dev) / 8).bfloat16()
+        us = timeit(lambda: matmul_nt(a, w), reps)
+        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
+        print(f"rocps gemm_bf16 {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
+              flush=True)
+        us = timeit(lambda: a @ w.t(), reps)
+        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
+        print(f"rocblas matmul  {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
+              flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--reps", type=int, default=50)
@@ -128,19 +144,3 @@ def main():
 
 if __name__ == "__main__":
     main()
-
-
-def bench_gemm(dev, reps):
-    from rocps.ops.gemm import matmul_nt
-    for (M, N, K) in [(4096, 4096, 4096), (8192, 768, 768),
-                      (8192, 3072, 768), (8192, 768, 3072)]:
-        a = (torch.randn(M, K, device=dev) / 8).bfloat16()
-        w = (torch.randn(N, K, device=dev) / 8).bfloat16()
-        us = timeit(lambda: matmul_nt(a, w), reps)
-        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
-        print(f"rocps gemm_bf16 {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
-              flush=True)
-        us = timeit(lambda: a @ w.t(), reps)
-        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
-        print(f"rocblas matmul  {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
-              flush=True)
