@@ -35,6 +35,8 @@ def parse_args():
                              "mnist_mlp"])
     ap.add_argument("--mode", type=str, default="sync",
                     choices=["sync", "async"])
+    ap.add_argument("--num-ps", type=int, default=2,
+                    help="async mode: host-DRAM PS shards (configs[2])")
     ap.add_argument("--bucket-mb", type=float, default=25.0)
     ap.add_argument("--seq-len", type=int, default=128)
     ap.add_argument("--dtype", type=str, default="bf16",
@@ -115,8 +117,42 @@ def main():
         dtype = torch.float32  # CPU sanity runs in fp32
 
     model, opt, make_batch = build(args, device, dtype)
-    engine = SyncEngine(model, comm, bucket_cap_mb=args.bucket_mb)
-    engine.broadcast_parameters(src=0)
+    ps_procs = []
+    if args.mode == "async":
+        # configs[2]: N GPU workers + --num-ps host-DRAM PS shards on this
+        # node. PS tasks are spawned as subprocesses by rank 0; gradient
+        # flow goes via shared-memory push/apply/pull; the torchrun group
+        # is used only for the timing barriers.
+        import subprocess
+        import sys as _sys
+
+        from rocps.cluster import ClusterSpec, ControlClient
+        from rocps.engine.async_engine import AsyncWorkerEngine
+
+        base = int(os.environ.get("MASTER_PORT", "29500"))
+        spec = {
+            "ps": [f"127.0.0.1:{base + 101 + i}" for i in range(args.num_ps)],
+            "worker": [f"127.0.0.1:{base + 201 + r}"
+                       for r in range(comm.world_size)],
+        }
+        cluster = ClusterSpec(spec)
+        if comm.rank == 0:
+            env = dict(os.environ, CUDA_VISIBLE_DEVICES="",
+                       HIP_VISIBLE_DEVICES="")
+            for i in range(args.num_ps):
+                ps_procs.append(subprocess.Popen(
+                    [_sys.executable, "-m", "rocps.cli.ps_server",
+                     "--cluster-json", json.dumps(spec),
+                     "--task-index", str(i)], env=env))
+        from rocps.cluster import Server
+        server = Server(cluster, "worker", comm.rank)
+        control = ControlClient(cluster.master_address)
+        engine = AsyncWorkerEngine(model, cluster, control, comm.rank,
+                                   device)
+        engine.initialize(opt)
+    else:
+        engine = SyncEngine(model, comm, bucket_cap_mb=args.bucket_mb)
+        engine.broadcast_parameters(src=0)
 
     n_batches = 4
     batches = [make_batch(1000 * comm.rank + i) for i in range(n_batches)]
@@ -227,6 +263,14 @@ def main():
             },
         }
         print(json.dumps(out), flush=True)
+    if args.mode == "async" and comm.rank == 0:
+        from rocps.cluster import shutdown_cluster
+        shutdown_cluster(cluster)
+        for p in ps_procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                p.terminate()
     if comm.is_distributed:
         comm.destroy()
 

@@ -1,8 +1,9 @@
 // rocps._hipops — torch bindings for the hand-written CDNA4 kernels.
 //
 // Every function validates device/dtype/layout and launches on the
-// current HIP stream; per-channel BN folds run as C-sized torch ops on
-// device (L2-resident, negligible next to the M×C passes).
+// current HIP stream. All per-channel/per-row small math runs in
+// dedicated finalize kernels — host-side C-sized torch ops made the
+// ResNet step CPU-bound (profiles/r50_b256_torch_profile_hostbound_before.txt).
 
 #include <torch/extension.h>
 
@@ -402,6 +403,54 @@ at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
 }
 
 // ---------------------------------------------------------------------------
+// maxpool 3x3/s2/p1 (C9 pooling slice)
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> maxpool332_fwd(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4
+              && x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+      W = (int)x.size(3);
+  TORCH_CHECK(C % 8 == 0, "maxpool332: C % 8 == 0");
+  int OH = (H + 1) / 2, OW = (W + 1) / 2;
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
+  auto y = at::empty({N, C, OH, OW},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = at::empty({N, OH, OW, C}, x.options().dtype(at::kByte));
+  auto st = cur_stream(x);
+  if (x.scalar_type() == at::kBFloat16)
+    launch_maxpool332_fwd_bf16(bf16_ptr(x), bf16_ptr(y),
+                               idx.data_ptr<unsigned char>(), N, C, H, W, OH,
+                               OW, st);
+  else
+    launch_maxpool332_fwd_f32(x.data_ptr<float>(), y.data_ptr<float>(),
+                              idx.data_ptr<unsigned char>(), N, C, H, W, OH,
+                              OW, st);
+  return {y, idx};
+}
+
+at::Tensor maxpool332_bwd(at::Tensor dy, at::Tensor idx, long H, long W) {
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4);
+  auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  int N = (int)dyc.size(0), C = (int)dyc.size(1), OH = (int)dyc.size(2),
+      OW = (int)dyc.size(3);
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(dy.device());
+  auto dx = at::empty({N, C, H, W},
+                      dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto st = cur_stream(dyc);
+  if (dyc.scalar_type() == at::kBFloat16)
+    launch_maxpool332_bwd_bf16(bf16_ptr(dyc), idx.data_ptr<unsigned char>(),
+                               bf16_ptr(dx), N, C, (int)H, (int)W, OH, OW,
+                               st);
+  else
+    launch_maxpool332_bwd_f32(dyc.data_ptr<float>(),
+                              idx.data_ptr<unsigned char>(),
+                              dx.data_ptr<float>(), N, C, (int)H, (int)W, OH,
+                              OW, st);
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "rocps hand-written CDNA4 (gfx950) kernels";
@@ -415,4 +464,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_fwd", &ln_fwd);
   m.def("ln_bwd", &ln_bwd);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("maxpool332_fwd", &maxpool332_fwd);
+  m.def("maxpool332_bwd", &maxpool332_bwd);
 }

@@ -114,3 +114,17 @@ void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
 void launch_gemm_bf16(const rbf16* A, const rbf16* W, const float* bias,
                       rbf16* D, int M, int N, int K, int epilogue,
                       hipStream_t st);
+
+// maxpool.hip (3x3 stride-2 pad-1, NHWC)
+void launch_maxpool332_fwd_f32(const float* x, float* y, unsigned char* idx,
+                               int N, int C, int H, int W, int OH, int OW,
+                               hipStream_t st);
+void launch_maxpool332_fwd_bf16(const rbf16* x, rbf16* y, unsigned char* idx,
+                                int N, int C, int H, int W, int OH, int OW,
+                                hipStream_t st);
+void launch_maxpool332_bwd_f32(const float* dy, const unsigned char* idx,
+                               float* dx, int N, int C, int H, int W, int OH,
+                               int OW, hipStream_t st);
+void launch_maxpool332_bwd_bf16(const rbf16* dy, const unsigned char* idx,
+                                rbf16* dx, int N, int C, int H, int W,
+                                int OH, int OW, hipStream_t st);

@@ -21,6 +21,7 @@ import torch
 import torch.nn as nn
 
 from rocps.ops.norm import BatchNormAct2d
+from rocps.ops.pool import MaxPool332
 
 __all__ = ["ResNet", "resnet50", "wide_resnet101_2"]
 
@@ -67,7 +68,7 @@ class ResNet(nn.Module):
         self.base_width = base_width
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = BatchNormAct2d(64, act="relu")
-        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.maxpool = MaxPool332()
         self.layer1 = self._make_layer(64, layers[0])
         self.layer2 = self._make_layer(128, layers[1], stride=2)
         self.layer3 = self._make_layer(256, layers[2], stride=2)

@@ -71,17 +71,18 @@ class Optimizer:
                           for k, st in self._state.items()}}
 
     def load_state_dict(self, sd: Dict[str, Any]) -> None:
+        """Restore lr + any already-materialised slot state.
+
+        Checkpoint restore of optimizer state goes through
+        ``Trainer._load_optimizer_state`` (which materialises slots
+        first); this method only covers same-process round trips.
+        """
         self.lr = sd["lr"]
         by_repr = {repr(k): k for k in self._state}
         for ks, st in sd["state"].items():
             if ks in by_repr:
                 for n, v in st.items():
                     self._state[by_repr[ks]][n].copy_(v)
-            else:
-                # state restored before first apply: stash under string key;
-                # slot_state() will not see it, so restore must happen after
-                # a first step OR callers pre-create state via warm_slots.
-                self._state[ks] = st  # type: ignore[index]
 
 
 class GradientDescentOptimizer(Optimizer):

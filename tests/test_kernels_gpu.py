@@ -404,3 +404,39 @@ def test_linear_gelu_autograd():
     assert torch.allclose(y.float(), y2, atol=5e-2, rtol=5e-2)
     assert torch.allclose(x.grad.float(), x2.grad, atol=5e-2, rtol=5e-2)
     assert torch.allclose(w.grad.float(), w2.grad, atol=5e-2, rtol=5e-2)
+
+
+# ---------------------------------------------------------------------------
+# maxpool 3x3/s2/p1 (C9 pooling)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("shape", [(8, 64, 112, 112), (4, 32, 15, 15)])
+def test_maxpool332_fp32_vs_torch(shape):
+    from rocps.ops.pool import MaxPool332
+    torch.manual_seed(8)
+    N, C, H, W = shape
+    # fp32 random: exact-tie probability negligible -> grads comparable
+    x = (torch.randn(N, C, H, W, device=DEV)
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    mp = MaxPool332()
+    y = mp(x)
+    ref_x = x.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.max_pool2d(ref_x, 3, stride=2, padding=1)
+    assert y.shape == ref.shape
+    assert torch.equal(y, ref)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g)
+    torch.cuda.synchronize()
+    assert torch.allclose(x.grad, ref_x.grad, atol=1e-6), \
+        (x.grad - ref_x.grad).abs().max().item()
+
+
+def test_maxpool332_bf16_forward_exact():
+    from rocps.ops.pool import MaxPool332
+    torch.manual_seed(9)
+    x = (torch.randn(4, 64, 56, 56, device=DEV).bfloat16()
+         .to(memory_format=torch.channels_last))
+    y = MaxPool332()(x)
+    ref = torch.nn.functional.max_pool2d(x, 3, stride=2, padding=1)
+    assert torch.equal(y, ref)
