@@ -292,6 +292,10 @@ class AsyncWorkerEngine:
             self.flat_params.append(fp)
             self.flat_grads.append(fg)
         self.segs: List[Optional[_ShmSeg]] = [None] * self.num_shards
+        # side stream: push/pull copies run off the compute stream (C7's
+        # third stream) so pinned DMA overlaps across shards
+        self._side = (torch.cuda.Stream(device=device)
+                      if device.type == "cuda" else None)
         self.local_step = 0
         self.is_chief = task_index == 0
         self._attached = False
@@ -375,10 +379,22 @@ class AsyncWorkerEngine:
         self.local_step += 1
         w = self.task_index
         # push grads to every shard, bump seq (D2H straight into the
-        # shard slot — pinned DMA when _pin_segment succeeded)
-        for s, seg in enumerate(self.segs):
-            seg.grads[w].copy_(self.flat_grads[s])
-            seg.header[8 + 2 * w] = self.local_step
+        # shard slot — pinned DMA on the side stream when _pin_segment
+        # succeeded)
+        if self._side is not None:
+            self._side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._side):
+                for s, seg in enumerate(self.segs):
+                    seg.grads[w].copy_(self.flat_grads[s],
+                                       non_blocking=getattr(seg, "pinned",
+                                                            False))
+            self._side.synchronize()  # grads visible before the seq bump
+            for s, seg in enumerate(self.segs):
+                seg.header[8 + 2 * w] = self.local_step
+        else:
+            for s, seg in enumerate(self.segs):
+                seg.grads[w].copy_(self.flat_grads[s])
+                seg.header[8 + 2 * w] = self.local_step
         # wait for own applies (the reference's RunGraph round trip)
         deadline = time.time() + 120.0
         for seg in self.segs:
@@ -391,8 +407,16 @@ class AsyncWorkerEngine:
         self.last_global_step = int(self.segs[0].header[5])
 
     def pull(self) -> None:
-        for s, seg in enumerate(self.segs):
-            self.flat_params[s].copy_(seg.params)
+        if self._side is not None:
+            with torch.cuda.stream(self._side):
+                for s, seg in enumerate(self.segs):
+                    self.flat_params[s].copy_(seg.params,
+                                              non_blocking=getattr(
+                                                  seg, "pinned", False))
+            torch.cuda.current_stream().wait_stream(self._side)
+        else:
+            for s, seg in enumerate(self.segs):
+                self.flat_params[s].copy_(seg.params)
 
     def detach(self) -> None:
         for seg in self.segs:
