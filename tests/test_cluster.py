@@ -122,3 +122,22 @@ def test_worker_server_dispatch(free_ports):
     finally:
         master.stop()
         worker.stop()
+
+
+def test_create_local_cluster():
+    """tf.test.create_local_cluster parity: in-process multi-server."""
+    from rocps.testing import create_local_cluster
+    workers, ps, cluster = create_local_cluster(num_workers=2, num_ps=1)
+    try:
+        assert cluster.num_tasks("worker") == 2
+        assert cluster.num_tasks("ps") == 1
+        c = ControlClient(cluster.master_address)
+        assert c.incr("k") == 1
+        for s in workers:
+            cc = ControlClient(f"127.0.0.1:{s.bound_port}")
+            assert cc.call("ping")["ok"]
+            cc.close()
+        c.close()
+    finally:
+        for s in workers + ps:
+            s.stop()
