@@ -29,10 +29,8 @@ path; workers attach.
 
 from __future__ import annotations
 
-import json
 import mmap
 import os
-import struct
 import threading
 import time
 from typing import Dict, List, Optional

@@ -10,7 +10,6 @@ softmax-xent kernel (vocab 30522).
 
 from __future__ import annotations
 
-import math
 
 import torch
 import torch.nn as nn

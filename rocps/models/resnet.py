@@ -15,7 +15,7 @@ MI355X-first choices:
 
 from __future__ import annotations
 
-from typing import List, Optional, Type
+from typing import List, Optional
 
 import torch
 import torch.nn as nn

@@ -16,7 +16,7 @@ grid-stride, one launch per bucket instead of one per variable.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Hashable, Optional
+from typing import Any, Dict, Hashable
 
 import torch
 
