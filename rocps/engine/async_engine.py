@@ -173,10 +173,23 @@ class AsyncPSShard:
                 target=self._apply_loop, daemon=True,
                 name=f"rocps-ps-apply-{self.shard_index}")
             self._apply_thread.start()
+            import atexit
+            atexit.register(self._cleanup)
         self.master.put(f"async_shm_{self.shard_index}",
                         {"path": path, "numel": numel,
                          "dtype": head["dtype"]})
         return {"ok": True, "path": path}, b""
+
+    def _cleanup(self) -> None:
+        """Remove the /dev/shm segment on PS shutdown (workers hold
+        their own mappings; the path must not leak on long-lived
+        nodes)."""
+        seg = self.seg
+        if seg is not None:
+            try:
+                os.unlink(seg.path)
+            except OSError:
+                pass
 
     def _on_stats(self, head, payload):
         return {"ok": True, "applied": self.applied,

@@ -144,3 +144,28 @@ def test_event_file_writer(tmp_path):
         off += 16 + ln
         n += 1
     assert n == 6  # file_version + 5 scalars
+
+
+def test_resnet_tiny_checkpoint_roundtrip(tmp_path):
+    """ResNet checkpoint: conv weights + BN params/buffers (incl. the
+    int64 num_batches_tracked) through the TF-V2 bundle."""
+    from rocps.models.resnet import ResNet
+    torch.manual_seed(21)
+    m = ResNet([1, 1, 1, 1], num_classes=10)
+    tr = Trainer(m, MomentumOptimizer(0.05, 0.9))
+    g = torch.Generator().manual_seed(0)
+    for _ in range(2):
+        x = torch.randn(2, 3, 64, 64, generator=g)
+        y = torch.randint(0, 10, (2,), generator=g)
+        tr.train_step(lambda: m.loss(x, y))
+    saver = Saver(tr)
+    prefix = saver.save(None, str(tmp_path / "r.ckpt"), global_step=2)
+
+    torch.manual_seed(99)
+    m2 = ResNet([1, 1, 1, 1], num_classes=10)
+    tr2 = Trainer(m2, MomentumOptimizer(0.05, 0.9))
+    Saver(tr2).restore(tr2, prefix)
+    assert tr2.global_step == 2
+    sd1, sd2 = m.state_dict(), m2.state_dict()
+    for k in sd1:
+        assert torch.equal(sd1[k], sd2[k]), k
