@@ -170,3 +170,43 @@ def test_async_worker_death_tolerated(free_ports, tmp_path):
         except AssertionError as e:
             last_err = e
     raise last_err
+
+
+def test_async_inprocess_cluster():
+    """In-process async PS (create_local_cluster): push/apply/pull via
+    shared memory, PS stats, overfit on a fixed batch."""
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    workers, ps, cluster = create_local_cluster(num_workers=1, num_ps=1)
+    try:
+        shard = AsyncPSShard.serve_on(ps[0], cluster)
+        torch.manual_seed(5)
+        m = mnist_mlp(hidden=16)
+        control = ControlClient(cluster.master_address)
+        eng = AsyncWorkerEngine(m, cluster, control, task_index=0,
+                                device=torch.device("cpu"))
+        opt = GradientDescentOptimizer(0.5)
+        eng.initialize(opt)
+        g = torch.Generator().manual_seed(1)
+        x, y = synthetic_mnist_batch(64, generator=g)
+        losses = []
+        for _ in range(25):
+            eng.zero_grad()
+            loss = m.loss(x, y)
+            eng.step(loss, opt)
+            losses.append(float(loss.detach()))
+        assert losses[-1] < losses[0]
+        assert shard.applied == 25
+        assert eng.last_global_step == 25
+        r = ControlClient(cluster.task_address("ps", 0)).call("shard_stats")
+        assert r["applied"] == 25 and r["dropped_stale"] == 0
+        eng.detach()
+    finally:
+        for s in workers + ps:
+            s.stop()
