@@ -94,3 +94,10 @@ __host__ __forceinline__ int rocps_grid_for(long work_items, int block,
              __FILE__, __LINE__);                                    \
     }                                                                \
   } while (0)
+
+__device__ __forceinline__ float wave_allreduce_sum(float v) {
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, WAVE);
+  return v;
+}
