@@ -61,15 +61,20 @@ __global__ void ln_fwd_vec_kernel(const T* __restrict__ x,
   const T* xr = x + row * D;
   T* yr = y + row * D;
   int nvec = D >> 3;
+  // two STATICALLY-indexed chunks per lane (D <= 1024; dispatch falls
+  // back to the block-per-row kernel beyond) — a runtime-indexed
+  // register array would spill to scratch (guide rule 20).
+  const bool has0 = lane < nvec;
+  const bool has1 = lane + WAVE < nvec;
+  float v0[8] = {0.f}, v1[8] = {0.f};
+  if (has0) lnload8<T>(xr + lane * 8, v0);
+  if (has1) lnload8<T>(xr + (lane + WAVE) * 8, v1);
   float s = 0.f, ss = 0.f;
-  for (int v8 = lane; v8 < nvec; v8 += WAVE) {
-    float v[8];
-    lnload8<T>(xr + v8 * 8, v);
 #pragma unroll
-    for (int k = 0; k < 8; k++) {
-      s += v[k];
-      ss = fmaf(v[k], v[k], ss);
-    }
+  for (int k = 0; k < 8; k++) {
+    s += v0[k] + v1[k];
+    ss = fmaf(v0[k], v0[k], ss);
+    ss = fmaf(v1[k], v1[k], ss);
   }
   s = wave_allreduce_sum(s);
   ss = wave_allreduce_sum(ss);
@@ -80,13 +85,19 @@ __global__ void ln_fwd_vec_kernel(const T* __restrict__ x,
     mean[row] = mu;
     invstd[row] = is;
   }
-  for (int v8 = lane; v8 < nvec; v8 += WAVE) {
-    float v[8];
-    lnload8<T>(xr + v8 * 8, v);
+  if (has0) {
 #pragma unroll
     for (int k = 0; k < 8; k++)
-      v[k] = fmaf((v[k] - mu) * is, gamma[v8 * 8 + k], beta[v8 * 8 + k]);
-    lnstore8<T>(yr + v8 * 8, v);
+      v0[k] = fmaf((v0[k] - mu) * is, gamma[lane * 8 + k],
+                   beta[lane * 8 + k]);
+    lnstore8<T>(yr + lane * 8, v0);
+  }
+  if (has1) {
+#pragma unroll
+    for (int k = 0; k < 8; k++)
+      v1[k] = fmaf((v1[k] - mu) * is, gamma[(lane + WAVE) * 8 + k],
+                   beta[(lane + WAVE) * 8 + k]);
+    lnstore8<T>(yr + (lane + WAVE) * 8, v1);
   }
 }
 
@@ -106,31 +117,49 @@ __global__ void ln_bwd_dx_vec_kernel(const T* __restrict__ x,
   T* dxr = dx + row * D;
   float mu = mean[row], is = invstd[row];
   int nvec = D >> 3;
-  float c1 = 0.f, c2 = 0.f;
-  for (int v8 = lane; v8 < nvec; v8 += WAVE) {
-    float xv[8], dv[8];
-    lnload8<T>(xr + v8 * 8, xv);
-    lnload8<T>(dyr + v8 * 8, dv);
+  const bool has0 = lane < nvec;
+  const bool has1 = lane + WAVE < nvec;
+  float x0[8] = {0.f}, x1[8] = {0.f}, d0[8] = {0.f}, d1[8] = {0.f};
+  float g0[8] = {0.f}, g1[8] = {0.f};
+  if (has0) {
+    lnload8<T>(xr + lane * 8, x0);
+    lnload8<T>(dyr + lane * 8, d0);
 #pragma unroll
-    for (int k = 0; k < 8; k++) {
-      float g = dv[k] * gamma[v8 * 8 + k];
-      c1 = fmaf(g, (xv[k] - mu) * is, c1);
-      c2 += g;
-    }
+    for (int k = 0; k < 8; k++) g0[k] = gamma[lane * 8 + k];
+  }
+  if (has1) {
+    lnload8<T>(xr + (lane + WAVE) * 8, x1);
+    lnload8<T>(dyr + (lane + WAVE) * 8, d1);
+#pragma unroll
+    for (int k = 0; k < 8; k++) g1[k] = gamma[(lane + WAVE) * 8 + k];
+  }
+  float c1 = 0.f, c2 = 0.f;
+#pragma unroll
+  for (int k = 0; k < 8; k++) {
+    float a0 = d0[k] * g0[k], a1 = d1[k] * g1[k];
+    c1 = fmaf(a0, (x0[k] - mu) * is, c1);
+    c1 = fmaf(a1, (x1[k] - mu) * is, c1);
+    c2 += a0 + a1;
   }
   c1 = wave_allreduce_sum(c1) / D;
   c2 = wave_allreduce_sum(c2) / D;
-  for (int v8 = lane; v8 < nvec; v8 += WAVE) {
-    float xv[8], dv[8];
-    lnload8<T>(xr + v8 * 8, xv);
-    lnload8<T>(dyr + v8 * 8, dv);
+  if (has0) {
 #pragma unroll
     for (int k = 0; k < 8; k++) {
-      float g = dv[k] * gamma[v8 * 8 + k];
-      float xh = (xv[k] - mu) * is;
-      dv[k] = is * (g - c2 - xh * c1);
+      float g = d0[k] * g0[k];
+      float xh = (x0[k] - mu) * is;
+      d0[k] = is * (g - c2 - xh * c1);
     }
-    lnstore8<T>(dxr + v8 * 8, dv);
+    lnstore8<T>(dxr + lane * 8, d0);
+  }
+  if (has1) {
+#pragma unroll
+    for (int k = 0; k < 8; k++) {
+      float g = d1[k] * g1[k];
+      float xh = (x1[k] - mu) * is;
+      d1[k] = is * (g - c2 - xh * c1);
+    }
+    lnstore8<T>(dxr + (lane + WAVE) * 8, d1);
   }
 }
 
@@ -246,7 +275,7 @@ static int ln_ct_for(int C) {
 void launch_ln_fwd_f32(const float* x, const float* gamma, const float* beta,
                        float* y, float* mean, float* invstd, long N, int D,
                        float eps, hipStream_t st) {
-  if (D % 8 == 0) {
+  if (D % 8 == 0 && D <= 1024) {
     unsigned grid = (unsigned)((N + 3) / 4);
     hipLaunchKernelGGL((ln_fwd_vec_kernel<float>), dim3(grid), dim3(256), 0,
                        st, x, gamma, beta, y, mean, invstd, N, D, eps);
@@ -258,7 +287,7 @@ void launch_ln_fwd_f32(const float* x, const float* gamma, const float* beta,
 void launch_ln_fwd_bf16(const rbf16* x, const float* gamma, const float* beta,
                         rbf16* y, float* mean, float* invstd, long N, int D,
                         float eps, hipStream_t st) {
-  if (D % 8 == 0) {
+  if (D % 8 == 0 && D <= 1024) {
     unsigned grid = (unsigned)((N + 3) / 4);
     hipLaunchKernelGGL((ln_fwd_vec_kernel<bf16>), dim3(grid), dim3(256), 0,
                        st, reinterpret_cast<const bf16*>(x), gamma, beta,
@@ -273,7 +302,7 @@ void launch_ln_fwd_bf16(const rbf16* x, const float* gamma, const float* beta,
 void launch_ln_bwd_dx_f32(const float* x, const float* dy, const float* gamma,
                           const float* mean, const float* invstd, float* dx,
                           long N, int D, hipStream_t st) {
-  if (D % 8 == 0) {
+  if (D % 8 == 0 && D <= 1024) {
     unsigned grid = (unsigned)((N + 3) / 4);
     hipLaunchKernelGGL((ln_bwd_dx_vec_kernel<float>), dim3(grid), dim3(256),
                        0, st, x, dy, gamma, mean, invstd, dx, N, D);
@@ -286,7 +315,7 @@ void launch_ln_bwd_dx_bf16(const rbf16* x, const rbf16* dy,
                            const float* gamma, const float* mean,
                            const float* invstd, rbf16* dx, long N, int D,
                            hipStream_t st) {
-  if (D % 8 == 0) {
+  if (D % 8 == 0 && D <= 1024) {
     unsigned grid = (unsigned)((N + 3) / 4);
     hipLaunchKernelGGL((ln_bwd_dx_vec_kernel<bf16>), dim3(grid), dim3(256),
                        0, st, reinterpret_cast<const bf16*>(x),
