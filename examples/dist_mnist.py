@@ -5,7 +5,7 @@ Run exactly like the reference (one command per process):
 
     python examples/dist_mnist.py --ps_hosts=127.0.0.1:2222 \
         --worker_hosts=127.0.0.1:2223,127.0.0.1:2224 \
-        --job_name=ps --job_name=ps --task_index=0
+        --job_name=ps --task_index=0
     python examples/dist_mnist.py ... --job_name=worker --task_index=0
     python examples/dist_mnist.py ... --job_name=worker --task_index=1 --issync=1
 

@@ -141,3 +141,57 @@ def test_create_local_cluster():
     finally:
         for s in workers + ps:
             s.stop()
+
+
+def test_control_plane_concurrency_stress(free_ports):
+    """Hammer the native server's kv/counter/barrier state from many
+    threads — catches data races in csrc/control/control_plane.cc."""
+    import threading
+    ports = free_ports(3)
+    cs = _mk_cluster(ports)
+    master = Server(cs, "ps", 0)
+    try:
+        n_threads, n_ops = 16, 60
+        errors = []
+
+        def worker(t):
+            try:
+                c = ControlClient(cs.master_address)
+                for i in range(n_ops):
+                    c.incr("shared")
+                    c.put(f"k{t}_{i}", {"t": t, "i": i})
+                    assert c.get(f"k{t}_{i}") == {"t": t, "i": i}
+                c.barrier("stress", f"t{t}", n_threads, timeout=60)
+                c.close()
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+
+        ts = [threading.Thread(target=worker, args=(t,))
+              for t in range(n_threads)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(90)
+        assert not errors, errors[:3]
+        c = ControlClient(cs.master_address)
+        assert c.read_counter("shared") == n_threads * n_ops
+        c.close()
+    finally:
+        master.stop()
+
+
+def test_pure_python_server_fallback(free_ports, monkeypatch):
+    """The protocol-identical Python server (ROCPS_PURE_PY_SERVER=1)."""
+    monkeypatch.setenv("ROCPS_PURE_PY_SERVER", "1")
+    ports = free_ports(3)
+    cs = _mk_cluster(ports)
+    master = Server(cs, "ps", 0)
+    try:
+        assert master._native is None  # really the python path
+        c = ControlClient(cs.master_address)
+        c.put("x", [1, 2, 3])
+        assert c.get("x") == [1, 2, 3]
+        assert c.incr("n") == 1
+        c.close()
+    finally:
+        master.stop()
