@@ -16,8 +16,11 @@ from rocps.train.session import (  # noqa: F401
 from rocps.train.hooks import (  # noqa: F401
     SessionRunHook, StopAtStepHook, StepCounterHook, NanTensorHook,
     CheckpointSaverHook, SummarySaverHook, LoggingTensorHook)
-from rocps.train.saver import Saver, latest_checkpoint  # noqa: F401
+from rocps.train.saver import (  # noqa: F401
+    Saver, latest_checkpoint, NewCheckpointReader)
 from rocps.train.coordinator import Coordinator, LooseVersionThread  # noqa: F401
 from rocps.train.global_step import (  # noqa: F401
     get_or_create_global_step, get_global_step)
 from rocps.train.supervisor import Supervisor  # noqa: F401
+from rocps.train.schedules import (  # noqa: F401
+    exponential_decay, cosine_decay, polynomial_decay, LearningRateHook)

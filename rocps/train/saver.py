@@ -15,7 +15,8 @@ from typing import Dict, List, Optional
 
 import torch
 
-__all__ = ["Saver", "latest_checkpoint", "get_checkpoint_state"]
+__all__ = ["Saver", "latest_checkpoint", "get_checkpoint_state",
+           "NewCheckpointReader"]
 
 
 def _state_file(checkpoint_dir: str) -> str:
@@ -119,3 +120,39 @@ class Saver:
     def last_checkpoints(checkpoint_dir: str) -> List[str]:
         st = get_checkpoint_state(checkpoint_dir)
         return list(st["all_model_checkpoint_paths"]) if st else []
+
+
+class NewCheckpointReader:
+    """``tf.train.NewCheckpointReader`` parity: inspect a TF-V2 bundle."""
+
+    def __init__(self, prefix: str):
+        from rocps.checkpoint.bundle import BundleReader, TORCH_DTYPE
+        self._r = BundleReader(prefix)
+        self._dtypes = TORCH_DTYPE
+
+    def get_variable_to_shape_map(self):
+        return {e["name"]: list(e["shape"]) for e in self._r.list_tensors()}
+
+    def get_variable_to_dtype_map(self):
+        return {e["name"]: self._dtypes[e["dtype"]]
+                for e in self._r.list_tensors()}
+
+    def has_tensor(self, name: str) -> bool:
+        return self._r.contains(name)
+
+    def get_tensor(self, name: str):
+        from rocps.checkpoint.bundle import read_bundle
+        return read_bundle_single(self._r, name, self._dtypes)
+
+
+def read_bundle_single(reader, name: str, dtypes):
+    import numpy as np
+    import torch as _torch
+    dtype_code, shape, _ = reader.entry_info(name)
+    dtype = dtypes[dtype_code]
+    raw = reader.read(name)
+    storage = {_torch.bfloat16: _torch.uint16,
+               _torch.bool: _torch.uint8}.get(dtype, dtype)
+    np_dtype = _torch.empty(0, dtype=storage).numpy().dtype
+    arr = np.frombuffer(raw, dtype=np_dtype).copy()
+    return _torch.from_numpy(arr).view(dtype).reshape(list(shape))

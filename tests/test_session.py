@@ -169,3 +169,33 @@ def test_resnet_tiny_checkpoint_roundtrip(tmp_path):
     sd1, sd2 = m.state_dict(), m2.state_dict()
     for k in sd1:
         assert torch.equal(sd1[k], sd2[k]), k
+
+
+def test_new_checkpoint_reader(tmp_path):
+    import rocps
+    from rocps.checkpoint.bundle import write_bundle
+    prefix = str(tmp_path / "m.ckpt-5")
+    t = {"a/w": torch.randn(3, 4), "b": torch.arange(6, dtype=torch.int64)}
+    write_bundle(prefix, t)
+    r = rocps.train.NewCheckpointReader(prefix)
+    assert r.get_variable_to_shape_map() == {"a/w": [3, 4], "b": [6]}
+    assert r.has_tensor("a/w") and not r.has_tensor("zzz")
+    assert torch.equal(r.get_tensor("b"), t["b"])
+    assert torch.equal(r.get_tensor("a/w"), t["a/w"])
+
+
+def test_lr_schedules_and_hook():
+    import rocps
+    lr = rocps.train.exponential_decay(0.1, 10, 0.5, staircase=True)
+    assert lr(0) == 0.1 and lr(10) == 0.05 and lr(25) == 0.025
+    lr2 = rocps.train.cosine_decay(1.0, 100)
+    assert abs(lr2(0) - 1.0) < 1e-9 and abs(lr2(100)) < 1e-9
+    lr3 = rocps.train.polynomial_decay(1.0, 100, end_learning_rate=0.1)
+    assert abs(lr3(100) - 0.1) < 1e-9
+
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    hook = rocps.train.LearningRateHook(
+        tr.optimizer, rocps.train.exponential_decay(0.1, 2, 0.5,
+                                                    staircase=True))
+    _run(tr, 5, hooks=[hook])
+    assert abs(tr.optimizer.lr - 0.1 * 0.5 ** 2) < 1e-12
