@@ -32,6 +32,9 @@ flags.DEFINE_string("worker_hosts", "", "comma-separated host:port list")
 flags.DEFINE_string("job_name", "", "'ps' or 'worker'")
 flags.DEFINE_integer("task_index", 0, "task index within the job")
 flags.DEFINE_boolean("issync", True, "synchronous (exactly-N) aggregation")
+flags.DEFINE_integer("replicas_to_aggregate", 0,
+                     "sync mode: aggregate K of N workers per step "
+                     "(0 = all; K<N = the reference's backup-worker mode)")
 flags.DEFINE_integer("train_steps", 50, "number of global steps")
 flags.DEFINE_integer("batch_size", 32, "per-worker batch size")
 flags.DEFINE_float("learning_rate", 0.1, "SGD learning rate")
@@ -52,8 +55,11 @@ def main(_):
     })
     server = rocps.train.Server(cluster, FLAGS.job_name, FLAGS.task_index)
 
+    nw = cluster.num_tasks("worker")
+    semisync = bool(FLAGS.issync and FLAGS.replicas_to_aggregate
+                    and FLAGS.replicas_to_aggregate < nw)
     if FLAGS.job_name == "ps":
-        if not FLAGS.issync:
+        if not FLAGS.issync or semisync:
             from rocps.engine.async_engine import AsyncPSShard
             AsyncPSShard.serve_on(server, cluster)
         server.join()
@@ -66,7 +72,9 @@ def main(_):
     opt = rocps.train.GradientDescentOptimizer(FLAGS.learning_rate)
     if FLAGS.issync:
         opt = rocps.train.SyncReplicasOptimizer(
-            opt, replicas_to_aggregate=num_workers,
+            opt,
+            replicas_to_aggregate=(FLAGS.replicas_to_aggregate
+                                   or num_workers),
             total_num_replicas=num_workers)
     trainer = rocps.train.Trainer(
         model, opt, cluster=cluster, job_name=FLAGS.job_name,

@@ -54,16 +54,17 @@ def _np_dtype(t: torch.dtype):
 
 
 class _ShmLayout:
-    """header int64[8 + 2*W] | params | grads[W]  (offsets 256-aligned).
+    """header int64[8 + 3*W] | params | grads[W]  (offsets 256-aligned).
 
     header: [magic, version(applies), num_workers, numel, dtype_code,
-             global_step, reserved, reserved, then per worker: seq, ack]
+             global_step, round, reserved, then per worker:
+             seq, ack, pver(param round the worker last pulled)]
     """
 
     def __init__(self, numel: int, num_workers: int, itemsize: int):
         self.numel = numel
         self.num_workers = num_workers
-        self.header_words = 8 + 2 * num_workers
+        self.header_words = 8 + 3 * num_workers
         self.params_off = ((self.header_words * 8 + _ALIGN - 1)
                            // _ALIGN * _ALIGN)
         grad_bytes = numel * itemsize
@@ -74,6 +75,18 @@ class _ShmLayout:
 
     def grad_off(self, w: int) -> int:
         return self.grad_off0 + w * self.grad_stride
+
+    @staticmethod
+    def seq_idx(w: int) -> int:
+        return 8 + 3 * w
+
+    @staticmethod
+    def ack_idx(w: int) -> int:
+        return 8 + 3 * w + 1
+
+    @staticmethod
+    def pver_idx(w: int) -> int:
+        return 8 + 3 * w + 2
 
 
 class _ShmSeg:
@@ -137,6 +150,9 @@ class AsyncPSShard:
         self.applied = 0
         self.dropped_stale = 0
         self.staleness_bound: Optional[int] = None
+        self.replicas_to_aggregate: Optional[int] = None
+        self.flush_timeout = 2.0
+        self.partial_rounds = 0
 
     @staticmethod
     def serve_on(server: Server, cluster: ClusterSpec) -> "AsyncPSShard":
@@ -168,6 +184,8 @@ class AsyncPSShard:
         opt_spec = head.get("optimizer", {"type": "sgd", "lr": 0.01})
         self.optimizer = _build_optimizer(opt_spec)
         self.staleness_bound = head.get("staleness_bound")
+        self.replicas_to_aggregate = head.get("replicas_to_aggregate")
+        self.flush_timeout = float(head.get("flush_timeout") or 2.0)
         if self._apply_thread is None:
             self._apply_thread = threading.Thread(
                 target=self._apply_loop, daemon=True,
@@ -194,19 +212,28 @@ class AsyncPSShard:
     def _on_stats(self, head, payload):
         return {"ok": True, "applied": self.applied,
                 "dropped_stale": self.dropped_stale,
+                "partial_rounds": self.partial_rounds,
                 "version": int(self.seg.header[1]) if self.seg else 0}, b""
 
     # -- the apply loop --------------------------------------------------
     def _apply_loop(self) -> None:
+        if self.replicas_to_aggregate:
+            self._semisync_loop()
+        else:
+            self._async_loop()
+
+    def _async_loop(self) -> None:
+        """Reference async mode: every push applied immediately."""
         seg = self.seg
         h = seg.header
         W = self.num_workers
+        L = _ShmLayout
         with self.coord.stop_on_exception():
             while not (self.coord.should_stop()
                        or self.server.shutdown_requested()):
                 progressed = False
                 for w in range(W):
-                    seq, ack = int(h[8 + 2 * w]), int(h[8 + 2 * w + 1])
+                    seq, ack = int(h[L.seq_idx(w)]), int(h[L.ack_idx(w)])
                     if seq != ack:
                         version = int(h[1])
                         # seq encodes (worker local step); stale-drop rule
@@ -219,9 +246,79 @@ class AsyncPSShard:
                             self.applied += 1
                             h[1] = version + 1
                         h[5] += 1  # global_step mirror
-                        h[8 + 2 * w + 1] = seq  # ack
+                        h[L.ack_idx(w)] = seq  # ack
                         progressed = True
                 if not progressed:
+                    time.sleep(0.0002)
+
+    def _semisync_loop(self) -> None:
+        """SyncReplicasOptimizer semantics (C4, replicas_to_aggregate=K
+        <= total): accumulate exactly K FRESH gradients per round
+        (fresh = computed against this round's parameters, i.e. the
+        worker's pver equals the current round — the reference's
+        local_step >= global_step check), average, apply ONCE, bump the
+        round, and only then ack the contributors (the token-queue
+        release). Stale pushes are dropped but acked so the straggler
+        re-pulls and rejoins."""
+        seg = self.seg
+        h = seg.header
+        W = self.num_workers
+        K = self.replicas_to_aggregate
+        L = _ShmLayout
+        accum = torch.zeros(seg.params.numel(), dtype=torch.float32)
+        pending: list = []  # contributors awaiting the round apply
+        n_contrib = 0
+        last_contrib_t = time.time()
+
+        def apply_round(n):
+            nonlocal n_contrib, pending, last_contrib_t
+            accum.div_(n)
+            self.optimizer.apply_flat(seg.params, accum.to(seg.params.dtype),
+                                      slot="shard")
+            self.applied += 1
+            h[1] += 1
+            h[6] = int(h[6]) + 1  # new round
+            accum.zero_()
+            n_contrib = 0
+            for pw in pending:  # token-queue release
+                h[L.ack_idx(pw)] = int(h[L.seq_idx(pw)])
+            pending = []
+            last_contrib_t = time.time()
+
+        with self.coord.stop_on_exception():
+            while not (self.coord.should_stop()
+                       or self.server.shutdown_requested()):
+                progressed = False
+                rnd = int(h[6])
+                for w in range(W):
+                    if w in pending:  # already contributed; awaiting apply
+                        continue
+                    seq, ack = int(h[L.seq_idx(w)]), int(h[L.ack_idx(w)])
+                    if seq == ack:
+                        continue
+                    progressed = True
+                    h[5] += 1  # global_step counts every worker step
+                    if int(h[L.pver_idx(w)]) != rnd:
+                        self.dropped_stale += 1   # stale round
+                        h[L.ack_idx(w)] = seq     # release the straggler
+                        continue
+                    accum += seg.grads[w].float()
+                    pending.append(w)
+                    n_contrib += 1
+                    last_contrib_t = time.time()
+                    if n_contrib >= K:
+                        apply_round(n_contrib)
+                        break  # re-read the new round
+                if not progressed:
+                    # liveness: at end-of-training the final round may
+                    # never reach K contributors — flush it so pending
+                    # workers are released (counted in partial_rounds)
+                    if (pending
+                            and time.time() - last_contrib_t
+                            > self.flush_timeout):
+                        self.partial_rounds += 1
+                        apply_round(n_contrib)
+                        continue
                     time.sleep(0.0002)
 
 
@@ -270,7 +367,9 @@ class AsyncWorkerEngine:
                  control: ControlClient, task_index: int,
                  device: torch.device, num_shards: Optional[int] = None,
                  strategy: str = "greedy",
-                 staleness_bound: Optional[int] = None):
+                 staleness_bound: Optional[int] = None,
+                 replicas_to_aggregate: Optional[int] = None,
+                 flush_timeout: float = 2.0):
         self.model = model
         self.cluster = cluster
         self.control = control
@@ -278,6 +377,8 @@ class AsyncWorkerEngine:
         self.device = device
         self.num_shards = num_shards or cluster.num_tasks("ps")
         self.staleness_bound = staleness_bound
+        self.replicas_to_aggregate = replicas_to_aggregate
+        self.flush_timeout = flush_timeout
         named = [(n, p) for n, p in model.named_parameters()
                  if p.requires_grad]
         self.plan = plan_shards(named, self.num_shards, strategy=strategy)
@@ -329,7 +430,9 @@ class AsyncWorkerEngine:
                                numel=flat.numel(),
                                dtype=str(self.dtype).replace("torch.", ""),
                                optimizer=optimizer_spec(optimizer),
-                               staleness_bound=self.staleness_bound)
+                               staleness_bound=self.staleness_bound,
+                               replicas_to_aggregate=self.replicas_to_aggregate,
+                               flush_timeout=self.flush_timeout)
                     if r.get("ok"):
                         break
                     # PS may still be wiring its shard service (startup race)
@@ -401,15 +504,16 @@ class AsyncWorkerEngine:
                                                             False))
             self._side.synchronize()  # grads visible before the seq bump
             for s, seg in enumerate(self.segs):
-                seg.header[8 + 2 * w] = self.local_step
+                seg.header[_ShmLayout.seq_idx(w)] = self.local_step
         else:
             for s, seg in enumerate(self.segs):
                 seg.grads[w].copy_(self.flat_grads[s])
-                seg.header[8 + 2 * w] = self.local_step
-        # wait for own applies (the reference's RunGraph round trip)
+                seg.header[_ShmLayout.seq_idx(w)] = self.local_step
+        # wait for own applies (the reference's RunGraph round trip; in
+        # semi-sync mode this is the token-queue wait for the round)
         deadline = time.time() + 120.0
         for seg in self.segs:
-            while int(seg.header[8 + 2 * w + 1]) != self.local_step:
+            while int(seg.header[_ShmLayout.ack_idx(w)]) != self.local_step:
                 if time.time() > deadline:
                     raise TimeoutError("PS apply timeout (shard dead?)")
                 time.sleep(0.0002)
@@ -418,6 +522,8 @@ class AsyncWorkerEngine:
         self.last_global_step = int(self.segs[0].header[5])
 
     def pull(self) -> None:
+        w = self.task_index
+        rounds = [int(seg.header[6]) for seg in self.segs]
         if self._side is not None:
             with torch.cuda.stream(self._side):
                 for s, seg in enumerate(self.segs):
@@ -428,6 +534,10 @@ class AsyncWorkerEngine:
         else:
             for s, seg in enumerate(self.segs):
                 self.flat_params[s].copy_(seg.params)
+        # claim the round read BEFORE the copy: if an apply landed
+        # mid-pull our grads will be (conservatively) counted stale
+        for s, seg in enumerate(self.segs):
+            seg.header[_ShmLayout.pver_idx(w)] = rounds[s]
 
     def detach(self) -> None:
         for seg in self.segs:

@@ -54,6 +54,17 @@ class Trainer:
             if torch.cuda.is_available() else torch.device("cpu"))
         self.model = model.to(self.device)
         self.optimizer = optimizer
+        # the reference's backup-worker mode (SyncReplicasOptimizer with
+        # replicas_to_aggregate < total): exactly-K aggregation lives on
+        # the PS shard (semisync), not on collectives
+        from rocps.train.sync_replicas import SyncReplicasOptimizer
+        self.replicas_to_aggregate = None
+        if (mode == "sync" and isinstance(optimizer, SyncReplicasOptimizer)
+                and optimizer.replicas_to_aggregate
+                < optimizer.total_num_replicas
+                and cluster is not None and "ps" in cluster.jobs):
+            mode = "async"
+            self.replicas_to_aggregate = optimizer.replicas_to_aggregate
         if comm is None:
             if mode == "async":
                 # reference async mode has no worker collective group:
@@ -80,7 +91,8 @@ class Trainer:
             self.engine = AsyncWorkerEngine(
                 self.model, cluster=cluster, control=self.control,
                 task_index=task_index, device=self.device,
-                num_shards=num_ps_shards)
+                num_shards=num_ps_shards,
+                replicas_to_aggregate=self.replicas_to_aggregate)
         else:
             raise ValueError(f"unknown mode {mode!r}")
         self.global_step = 0

