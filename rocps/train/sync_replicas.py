@@ -8,11 +8,11 @@ the average, a token queue releases workers.
 MI355X recast: with ``replicas_to_aggregate == total_num_replicas``
 (the demo default) exactly-N aggregation + all-release is semantically
 a reduce-scatter → apply → all-gather barrier, which is what
-``SyncEngine`` runs; this wrapper therefore mostly *marks* the run as
-sync and carries the aggregation config. ``replicas_to_aggregate <
-total`` (backup-worker mode) keeps the reference's stale-drop
-bookkeeping: the engine checks the token's step tag and drops late
-contributions (async engine path).
+``SyncEngine`` runs. With ``replicas_to_aggregate < total``
+(backup-worker mode) the Trainer routes to the PS-shard *semisync*
+engine (rocps.engine.async_engine._semisync_loop): exactly-K fresh
+gradients per round, stale-drop with straggler release, token-queue
+semantics — the reference's ConditionalAccumulator behavior.
 """
 
 from __future__ import annotations
