@@ -44,6 +44,8 @@ def parse_args():
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--torch-profile", type=str, default=None,
                     help="write a torch.profiler summary of the timed steps")
+    ap.add_argument("--trace", type=str, default=None,
+                    help="write a Chrome trace JSON of the timed steps")
     ap.add_argument("--hip-graph", action="store_true",
                     help="capture the train step in a hipGraph and replay "
                          "(sync SGD/momentum only; removes launch gaps)")
@@ -204,7 +206,7 @@ def main():
     if use_cuda:
         torch.cuda.synchronize()
     prof = None
-    if args.torch_profile:
+    if args.torch_profile or args.trace:
         prof = torch.profiler.profile(
             activities=[torch.profiler.ProfilerActivity.CPU,
                         torch.profiler.ProfilerActivity.CUDA])
@@ -219,10 +221,13 @@ def main():
     if prof is not None:
         prof.__exit__(None, None, None)
         if comm.rank == 0:
-            with open(args.torch_profile, "w") as f:
-                f.write(prof.key_averages().table(
-                    sort_by="self_cuda_time_total" if use_cuda
-                    else "self_cpu_time_total", row_limit=60))
+            if args.torch_profile:
+                with open(args.torch_profile, "w") as f:
+                    f.write(prof.key_averages().table(
+                        sort_by="self_cuda_time_total" if use_cuda
+                        else "self_cpu_time_total", row_limit=60))
+            if args.trace:
+                prof.export_chrome_trace(args.trace)
 
     # max over ranks (conservative whole-job time)
     t = torch.tensor([elapsed], device=device if use_cuda else "cpu")

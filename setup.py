@@ -19,6 +19,11 @@ from setuptools import setup
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 os.environ.setdefault("MAX_JOBS", str(min(16, os.cpu_count() or 8)))
 
+# ROCPS_SANITIZE=address|thread builds the CPU extensions (bundle codec,
+# control plane) under the sanitizer — SURVEY §5 race-detection plan.
+_SAN = os.environ.get("ROCPS_SANITIZE")
+_SAN_FLAGS = [f"-fsanitize={_SAN}", "-fno-omit-frame-pointer"] if _SAN else []
+
 from torch.utils.cpp_extension import BuildExtension, CppExtension  # noqa: E402
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
@@ -27,12 +32,14 @@ ext_modules = [
     CppExtension(
         name="rocps._bundle",
         sources=["csrc/bundle/bundle.cc"],
-        extra_compile_args=["-O2", "-std=c++17"],
+        extra_compile_args=["-O2", "-std=c++17"] + _SAN_FLAGS,
+        extra_link_args=_SAN_FLAGS,
     ),
     CppExtension(
         name="rocps._ccp",
         sources=["csrc/control/control_plane.cc"],
-        extra_compile_args=["-O2", "-std=c++17"],
+        extra_compile_args=["-O2", "-std=c++17"] + _SAN_FLAGS,
+        extra_link_args=_SAN_FLAGS,
     ),
 ]
 
