@@ -164,10 +164,26 @@ class AsyncPSShard:
         numel = int(head["numel"])
         dtype = getattr(torch, head["dtype"])
         itemsize = torch.empty(0, dtype=dtype).element_size()
+        if self.seg is not None:
+            # re-init (chief restart): the apply thread still maps the
+            # segment, so never close/replace it — refresh params in
+            # place when the shape matches, reject otherwise.
+            if (self.seg.layout.numel != numel
+                    or self.seg.dtype != dtype):
+                return ({"ok": False,
+                         "error": "shard already initialized with a "
+                                  "different shape"}, b"")
+            src = np.frombuffer(payload, dtype=_np_dtype(dtype), count=numel)
+            t = torch.from_numpy(src.copy())
+            if dtype == torch.bfloat16:
+                t = t.view(torch.bfloat16)
+            self.seg.params.copy_(t)
+            self.master.put(f"async_shm_{self.shard_index}",
+                            {"path": self.seg.path, "numel": numel,
+                             "dtype": head["dtype"]})
+            return {"ok": True, "path": self.seg.path}, b""
         layout = _ShmLayout(numel, self.num_workers, itemsize)
         path = f"/dev/shm/rocps_shard{self.shard_index}_{os.getpid()}"
-        if self.seg is not None:
-            self.seg.close()
         self.seg = _ShmSeg(path, layout, dtype, create=True)
         h = self.seg.header
         h.zero_()
