@@ -265,14 +265,20 @@ class ControlPlaneServer {
     }
     cv_.notify_all();
     if (accept_thread_.joinable()) accept_thread_.join();
-    std::vector<std::thread> conns;
-    {
-      std::lock_guard<std::mutex> g(conn_mu_);
-      conns.swap(conn_threads_);
-      for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+    // sweep until empty: a connection accepted in the instant before the
+    // listen socket closed may have spawned a thread after an earlier
+    // sweep — destroying a joinable std::thread calls std::terminate.
+    while (true) {
+      std::vector<std::thread> conns;
+      {
+        std::lock_guard<std::mutex> g(conn_mu_);
+        if (conn_threads_.empty()) break;
+        conns.swap(conn_threads_);
+        for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+      }
+      for (auto& t : conns)
+        if (t.joinable()) t.join();
     }
-    for (auto& t : conns)
-      if (t.joinable()) t.join();
   }
 
   bool shutdown_requested() {
@@ -303,6 +309,10 @@ class ControlPlaneServer {
       int one = 1;
       setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
       std::lock_guard<std::mutex> g(conn_mu_);
+      if (!running_) {  // raced with stop(): do not spawn past the sweep
+        ::close(fd);
+        break;
+      }
       conn_fds_.insert(fd);
       conn_threads_.emplace_back([this, fd] {
         conn_loop(fd);

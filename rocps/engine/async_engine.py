@@ -215,9 +215,14 @@ class AsyncPSShard:
         return {"ok": True, "path": path}, b""
 
     def _cleanup(self) -> None:
-        """Remove the /dev/shm segment on PS shutdown (workers hold
-        their own mappings; the path must not leak on long-lived
-        nodes)."""
+        """Shutdown path (atexit): stop + JOIN the apply thread first —
+        a daemon thread killed inside a torch op during interpreter
+        finalization aborts the process — then unlink the /dev/shm
+        segment (workers hold their own mappings)."""
+        self.coord.request_stop()
+        t = self._apply_thread
+        if t is not None and t.is_alive():
+            t.join(timeout=10.0)
         seg = self.seg
         if seg is not None:
             try:
