@@ -265,6 +265,7 @@ def main():
                 "seq_len": args.seq_len if args.model == "bert_base" else None,
                 "parallelism": f"dp{n}",
                 "mode": args.mode,
+                "num_ps": args.num_ps if args.mode == "async" else 0,
             },
         }
         print(json.dumps(out), flush=True)
