@@ -163,3 +163,26 @@ def test_index_structure_independent_parse(tmp_path):
     # raw data bytes are the LE float bytes at the recorded offset
     data = open(prefix + ".data-00000-of-00001", "rb").read()
     assert np.frombuffer(data[:24], dtype="<f4").tolist() == [0, 1, 2, 3, 4, 5]
+
+
+def test_edge_tensors_roundtrip(tmp_path):
+    """0-dim scalars, empty tensors, and single-element bf16."""
+    prefix = str(tmp_path / "e.ckpt")
+    t = {
+        "scalar0d": torch.tensor(3.25),
+        "empty": torch.empty(0, 4),
+        "one_bf16": torch.tensor([1.5]).bfloat16(),
+        "int64_scalar": torch.tensor(-7, dtype=torch.int64),
+    }
+    write_bundle(prefix, t)
+    back = read_bundle(prefix)
+    for k in t:
+        assert back[k].dtype == t[k].dtype, k
+        assert back[k].shape == t[k].shape, k
+        assert torch.equal(back[k], t[k]), k
+
+
+def test_empty_bundle(tmp_path):
+    prefix = str(tmp_path / "none.ckpt")
+    write_bundle(prefix, {})
+    assert read_bundle(prefix) == {}
