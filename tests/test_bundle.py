@@ -136,8 +136,7 @@ def test_index_structure_independent_parse(tmp_path):
     entries = _parse_block(raw[h_off:h_off + h_sz])
     keys = [k.decode() for k, _ in entries]
     assert keys == ["", "a/b", "z"]  # header first, then sorted
-    # block trailer crc over contents+type byte
-    import zlib  # noqa: F401  (zlib is crc32 (not c); use codec's helper)
+    # block trailer crc over contents+type byte (crc32c, not zlib crc32)
     block = raw[h_off:h_off + h_sz]
     typ = raw[h_off + h_sz:h_off + h_sz + 1]
     stored = struct.unpack("<I", raw[h_off + h_sz + 1:h_off + h_sz + 5])[0]
