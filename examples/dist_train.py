@@ -134,6 +134,10 @@ def main(_):
             _time.sleep(0.2)
         from rocps.cluster import shutdown_cluster
         shutdown_cluster(cluster)
+    # tear the collective group down explicitly: gloo's background
+    # threads racing interpreter exit abort with
+    # "terminate called without an active exception"
+    trainer.comm.destroy()
     return 0
 
 
