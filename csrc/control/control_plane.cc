@@ -20,6 +20,7 @@
 #include <pybind11/pybind11.h>
 
 #include <arpa/inet.h>
+#include <execinfo.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
@@ -502,6 +503,19 @@ class ControlPlaneServer {
 
 PYBIND11_MODULE(_ccp, m) {
   m.doc() = "rocps native TCP control plane (C1)";
+  // Debug aid: ROCPS_TERMINATE_BT=1 prints a native backtrace when
+  // std::terminate fires (e.g. a joinable std::thread destroyed), so
+  // exit-race aborts can be located without gdb.
+  if (const char* bt = std::getenv("ROCPS_TERMINATE_BT"); bt && bt[0] == '1') {
+    std::set_terminate([] {
+      void* frames[64];
+      int n = backtrace(frames, 64);
+      const char msg[] = "\n[rocps] std::terminate backtrace:\n";
+      (void)!write(2, msg, sizeof(msg) - 1);
+      backtrace_symbols_fd(frames, n, 2);
+      std::abort();
+    });
+  }
   py::class_<ControlPlaneServer>(m, "ControlPlaneServer")
       .def(py::init<int, bool, const std::string&, long>(), py::arg("port"),
            py::arg("is_master"), py::arg("job") = "", py::arg("task") = 0)

@@ -379,6 +379,15 @@ class Server:
         if self._native is not None:
             self._native.stop()
             self._native = None
+            # Join the watch thread before returning: it is blocked inside
+            # wait_shutdown() with the GIL released, and if the interpreter
+            # starts finalizing before it re-acquires the GIL, CPython
+            # pthread_exit()s it mid-pybind-frame -> std::terminate. stop()
+            # runs from atexit (interpreter still alive), so joining here
+            # guarantees the thread is gone before Py_Finalize.
+            w = getattr(self, "_watch", None)
+            if w is not None and w is not threading.current_thread():
+                w.join(timeout=10.0)
         if self._tcp is not None:
             self._tcp.shutdown()
             self._tcp.server_close()
