@@ -102,7 +102,8 @@ def main(argv: List[str] = None) -> int:
                 if r != 0:
                     print(f"[rocps-launch] {name} exited rc={r}; "
                           f"terminating cluster", flush=True)
-                    rc = r
+                    if rc == 0:  # first failure wins; peers we then
+                        rc = r   # SIGTERM exit -15 and must not mask it
                     for _, q in procs:
                         if q.poll() is None:
                             q.terminate()  # exact PID, never a pattern
