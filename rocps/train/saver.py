@@ -11,6 +11,7 @@ as the reference stack.
 from __future__ import annotations
 
 import os
+import time
 from typing import Dict, List, Optional
 
 import torch
@@ -54,12 +55,21 @@ def latest_checkpoint(checkpoint_dir: str) -> Optional[str]:
 
 
 class Saver:
-    def __init__(self, var_source=None, max_to_keep: int = 5):
+    def __init__(self, var_source=None, max_to_keep: int = 5,
+                 keep_checkpoint_every_n_hours: float = 10000.0):
         """``var_source``: a Trainer (state_tensors/load_state_tensors),
-        a dict of tensors, or None (pass tensors to save())."""
+        a dict of tensors, or None (pass tensors to save()).
+
+        ``keep_checkpoint_every_n_hours``: like the reference Saver —
+        a checkpoint that would be GC'd by ``max_to_keep`` is preserved
+        on disk (though dropped from the state file) if at least this
+        many hours passed since the last preserved one."""
         self._source = var_source
         self.max_to_keep = max_to_keep
+        self.keep_checkpoint_every_n_hours = keep_checkpoint_every_n_hours
         self._kept: List[str] = []
+        self._save_time: Dict[str, float] = {}
+        self._last_preserved: float = time.time()
 
     # ------------------------------------------------------------------
     def _collect(self, session_or_tensors) -> Dict[str, torch.Tensor]:
@@ -88,6 +98,7 @@ class Saver:
     def _update_state(self, prefix: str) -> None:
         d = os.path.dirname(prefix) or "."
         base = os.path.basename(prefix)
+        self._save_time[base] = time.time()
         self._kept = [p for p in self._kept if p != base] + [base]
         dropped, self._kept = (self._kept[:-self.max_to_keep],
                                self._kept[-self.max_to_keep:])
@@ -95,7 +106,12 @@ class Saver:
             f.write(f'model_checkpoint_path: "{base}"\n')
             for p in self._kept:
                 f.write(f'all_model_checkpoint_paths: "{p}"\n')
+        horizon = self.keep_checkpoint_every_n_hours * 3600.0
         for p in dropped:
+            ts = self._save_time.pop(p, 0.0)
+            if ts - self._last_preserved >= horizon:
+                self._last_preserved = ts  # preserved on disk, off-state
+                continue
             for suffix in (".index", ".data-00000-of-00001"):
                 try:
                     os.remove(os.path.join(d, p + suffix))

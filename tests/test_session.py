@@ -97,6 +97,23 @@ def test_max_to_keep_gc(tmp_path):
     assert Saver.last_checkpoints(ckdir) == ["model.ckpt-2", "model.ckpt-3"]
 
 
+def test_keep_checkpoint_every_n_hours(tmp_path):
+    """n=0: every GC'd checkpoint is preserved on disk (off the state
+    file); the default (10000h) deletes as usual — reference Saver
+    semantics."""
+    ckdir = str(tmp_path / "ck4")
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    saver = Saver(tr, max_to_keep=1, keep_checkpoint_every_n_hours=0.0)
+    for s in [1, 2, 3]:
+        tr.global_step = s
+        saver.save(None, os.path.join(ckdir, "model.ckpt"), global_step=s)
+    files = os.listdir(ckdir)
+    # dropped from rotation but preserved on disk
+    assert "model.ckpt-1.index" in files and "model.ckpt-2.index" in files
+    assert Saver.last_checkpoints(ckdir) == ["model.ckpt-3"]
+    assert latest_checkpoint(ckdir).endswith("model.ckpt-3")
+
+
 def test_step_counter_hook():
     tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
     sc = StepCounterHook(every_n_steps=2, batch_size=16)
