@@ -8,6 +8,7 @@ test_dist_mnist.py / test_session.py.
 """
 
 import json
+import os
 import struct
 import threading
 import time
@@ -217,6 +218,38 @@ def test_shm_seg_cross_mapping_roundtrip(tmp_path, dtype):
     finally:
         a.close()
         b.close()
+
+
+# ---------------------------------------------------------------------------
+# Comm world_size==1 fast path (C6)
+# ---------------------------------------------------------------------------
+
+def test_comm_world1_shortcircuits():
+    from rocps.engine.comm import Comm
+    c = Comm(rank=0, world_size=1, backend="gloo",
+             device=torch.device("cpu"))
+    assert not c.is_distributed
+    t = torch.arange(4.0)
+    assert c.all_reduce(t).wait()
+    out = torch.zeros(4)
+    c.reduce_scatter(out, t.clone()).wait()
+    assert torch.equal(out, t)
+    gathered = torch.zeros(4)
+    c.all_gather(gathered, out).wait()
+    assert torch.equal(gathered, t)
+    c.broadcast(t).wait()
+    c.barrier()
+    c.destroy()  # no group -> no-op
+
+
+def test_comm_tune_env(monkeypatch):
+    from rocps.engine.comm import Comm
+    monkeypatch.delenv("NCCL_MIN_NCHANNELS", raising=False)
+    monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    Comm.tune_env()
+    # >=8 channels so rings stripe across the 7 xGMI links
+    assert int(os.environ["NCCL_MIN_NCHANNELS"]) >= 8
+    assert os.environ["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
 
 
 # ---------------------------------------------------------------------------
