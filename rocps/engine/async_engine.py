@@ -165,23 +165,29 @@ class AsyncPSShard:
         dtype = getattr(torch, head["dtype"])
         itemsize = torch.empty(0, dtype=dtype).element_size()
         if self.seg is not None:
-            # re-init (chief restart): the apply thread still maps the
-            # segment, so never close/replace it — refresh params in
-            # place when the shape matches, reject otherwise.
+            # Re-init (chief restart): the apply thread still maps the
+            # segment, so never close/replace it. Reference semantics
+            # (SessionManager.prepare_session): an already-initialized
+            # model is ADOPTED by a restarting chief — its fresh random
+            # weights must not clobber PS progress — unless the chief
+            # explicitly restored a checkpoint (``force``), which wins.
             if (self.seg.layout.numel != numel
                     or self.seg.dtype != dtype):
                 return ({"ok": False,
                          "error": "shard already initialized with a "
                                   "different shape"}, b"")
-            src = np.frombuffer(payload, dtype=_np_dtype(dtype), count=numel)
-            t = torch.from_numpy(src.copy())
-            if dtype == torch.bfloat16:
-                t = t.view(torch.bfloat16)
-            self.seg.params.copy_(t)
+            if head.get("force"):
+                src = np.frombuffer(payload, dtype=_np_dtype(dtype),
+                                    count=numel)
+                t = torch.from_numpy(src.copy())
+                if dtype == torch.bfloat16:
+                    t = t.view(torch.bfloat16)
+                self.seg.params.copy_(t)
             self.master.put(f"async_shm_{self.shard_index}",
                             {"path": self.seg.path, "numel": numel,
                              "dtype": head["dtype"]})
-            return {"ok": True, "path": self.seg.path}, b""
+            return {"ok": True, "path": self.seg.path,
+                    "adopted": not head.get("force")}, b""
         layout = _ShmLayout(numel, self.num_workers, itemsize)
         path = f"/dev/shm/rocps_shard{self.shard_index}_{os.getpid()}"
         self.seg = _ShmSeg(path, layout, dtype, create=True)
@@ -435,7 +441,14 @@ class AsyncWorkerEngine:
         self.last_global_step = 0
 
     # -- bootstrap -------------------------------------------------------
-    def initialize(self, optimizer) -> None:
+    def initialize(self, optimizer, force_load: Optional[bool] = None) -> None:
+        """Chief: init_shard every PS shard. ``force_load`` (or the
+        ``force_init_load`` attribute, set by the session after a
+        checkpoint restore) makes the pushed params overwrite an
+        already-initialized shard; otherwise a re-init adopts the PS
+        state (reference SessionManager.prepare_session semantics)."""
+        force = (force_load if force_load is not None
+                 else getattr(self, "force_init_load", False))
         if self.is_chief:
             for s in range(self.num_shards):
                 ps_addr = self.cluster.task_address(
@@ -453,7 +466,8 @@ class AsyncWorkerEngine:
                                optimizer=optimizer_spec(optimizer),
                                staleness_bound=self.staleness_bound,
                                replicas_to_aggregate=self.replicas_to_aggregate,
-                               flush_timeout=self.flush_timeout)
+                               flush_timeout=self.flush_timeout,
+                               force=bool(force))
                     if r.get("ok"):
                         break
                     # PS may still be wiring its shard service (startup race)

@@ -236,6 +236,11 @@ class MonitoredTrainingSession:
             ckpt = latest_checkpoint(self.checkpoint_dir)
             if ckpt:
                 self.saver.restore(self.trainer, ckpt)
+                # async mode: the restored weights must overwrite an
+                # already-initialized PS shard (a plain chief restart
+                # without a checkpoint instead ADOPTS the PS state —
+                # reference SessionManager.prepare_session semantics)
+                self.trainer.engine.force_init_load = True
         self.trainer.sync_initial_state()
         for h in self._hooks:
             h.begin()

@@ -221,6 +221,70 @@ def test_async_inprocess_cluster():
             s.stop()
 
 
+def test_async_worker_restart_reattaches():
+    """A restarted worker (same task_index) re-runs init_shard; the
+    shard's safe re-init must preserve the PS parameters and step count
+    so training resumes instead of restarting from scratch."""
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    workers, ps, cluster = create_local_cluster(num_workers=1, num_ps=1)
+    try:
+        shard = AsyncPSShard.serve_on(ps[0], cluster)
+        g = torch.Generator().manual_seed(1)
+        x, y = synthetic_mnist_batch(64, generator=g)
+
+        def run_worker(n_steps):
+            torch.manual_seed(5)  # fresh process = same model init
+            m = mnist_mlp(hidden=16)
+            control = ControlClient(cluster.master_address)
+            eng = AsyncWorkerEngine(m, cluster, control, task_index=0,
+                                    device=torch.device("cpu"))
+            opt = GradientDescentOptimizer(0.5)
+            eng.initialize(opt)
+            losses = []
+            for _ in range(n_steps):
+                eng.zero_grad()
+                loss = m.loss(x, y)
+                eng.step(loss, opt)
+                losses.append(float(loss.detach()))
+            eng.detach()
+            return losses
+
+        first = run_worker(10)        # "crashes" after 10 steps
+        assert shard.applied == 10
+        second = run_worker(10)       # restarted replacement, same index
+        # PS progress survived the restart (adopt, not re-init): steps
+        # continued 10 -> 20 and the replacement resumed near where the
+        # first run left off, not back at the random-init loss.
+        assert shard.applied == 20
+        r = ControlClient(cluster.task_address("ps", 0)).call("shard_stats")
+        assert r["applied"] == 20
+        assert second[0] <= first[-1] + 1e-4, (first[-1], second[0])
+        assert second[0] < first[0] - 0.01  # NOT back at random init
+        assert second[-1] < second[0]
+        # checkpoint-restore path: force_init_load overwrites the shard,
+        # so training restarts from the (random-init) pushed weights
+        torch.manual_seed(5)
+        m3 = mnist_mlp(hidden=16)
+        eng3 = AsyncWorkerEngine(m3, cluster,
+                                 ControlClient(cluster.master_address),
+                                 task_index=0, device=torch.device("cpu"))
+        eng3.force_init_load = True
+        eng3.initialize(GradientDescentOptimizer(0.5))
+        loss3 = float(m3.loss(x, y).detach())
+        assert abs(loss3 - first[0]) < 1e-5, (loss3, first[0])
+        eng3.detach()
+    finally:
+        for s in workers + ps:
+            s.stop()
+
+
 def test_semisync_exactly_k_aggregation():
     """SyncReplicasOptimizer with replicas_to_aggregate=2 of 3 workers
     (the reference's backup-worker mode): exactly-K fresh gradients per
