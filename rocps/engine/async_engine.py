@@ -183,6 +183,7 @@ class AsyncPSShard:
                 if dtype == torch.bfloat16:
                     t = t.view(torch.bfloat16)
                 self.seg.params.copy_(t)
+                self.seg.header[5] = int(head.get("global_step") or 0)
             self.master.put(f"async_shm_{self.shard_index}",
                             {"path": self.seg.path, "numel": numel,
                              "dtype": head["dtype"]})
@@ -196,6 +197,9 @@ class AsyncPSShard:
         h[0] = _MAGIC
         h[2] = self.num_workers
         h[3] = numel
+        # restored run: the reference's global_step lives on the PS and
+        # is restored with the variables — continue counting from it
+        h[5] = int(head.get("global_step") or 0)
         # initial params from chief
         src = np.frombuffer(payload, dtype=_np_dtype(dtype), count=numel)
         t = torch.from_numpy(src.copy())
@@ -467,7 +471,9 @@ class AsyncWorkerEngine:
                                staleness_bound=self.staleness_bound,
                                replicas_to_aggregate=self.replicas_to_aggregate,
                                flush_timeout=self.flush_timeout,
-                               force=bool(force))
+                               force=bool(force),
+                               global_step=int(getattr(
+                                   self, "init_global_step", 0)))
                     if r.get("ok"):
                         break
                     # PS may still be wiring its shard service (startup race)

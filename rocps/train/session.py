@@ -239,8 +239,11 @@ class MonitoredTrainingSession:
                 # async mode: the restored weights must overwrite an
                 # already-initialized PS shard (a plain chief restart
                 # without a checkpoint instead ADOPTS the PS state —
-                # reference SessionManager.prepare_session semantics)
+                # reference SessionManager.prepare_session semantics);
+                # the restored global_step continues on the PS counter
                 self.trainer.engine.force_init_load = True
+                self.trainer.engine.init_global_step = \
+                    self.trainer.global_step
         self.trainer.sync_initial_state()
         for h in self._hooks:
             h.begin()

@@ -145,6 +145,30 @@ def test_async_two_workers_two_ps(free_ports, tmp_path):
     assert total_steps >= 40  # cluster-wide step target reached
 
 
+def test_async_checkpoint_resume(free_ports, tmp_path):
+    """Async-mode resume across cluster restarts: the chief restores the
+    checkpoint, force-loads it into the (fresh) PS shard together with
+    the restored global_step, and training continues 8 -> 14 instead of
+    restarting from 0."""
+    from rocps.train.saver import latest_checkpoint
+
+    ckdir = str(tmp_path / "ack")
+    p1, _ = _run_cluster(
+        free_ports, tmp_path / "p1", issync=False, n_workers=1, n_ps=1,
+        steps=8, batch=16,
+        extra=[f"--checkpoint_dir={ckdir}", "--save_checkpoint_steps=4"])
+    assert latest_checkpoint(ckdir).endswith("model.ckpt-8")
+    p2, _ = _run_cluster(
+        free_ports, tmp_path / "p2", issync=False, n_workers=1, n_ps=1,
+        steps=14, batch=16,
+        extra=[f"--checkpoint_dir={ckdir}", "--save_checkpoint_steps=4"])
+    # resumed at step 8: only ~6 further steps, counter continued to 14
+    assert latest_checkpoint(ckdir).endswith("model.ckpt-14")
+    assert len(p2[0]) <= 8  # not re-trained from scratch
+    # loss continued from the phase-1 weights, not from random init
+    assert p2[0][0] < p1[0][0] - 0.01, (p1[0][0], p2[0][0])
+
+
 def test_async_worker_death_tolerated(free_ports, tmp_path):
     """SURVEY §5 failure handling: in async mode a dead worker is
     tolerated — the PS keeps serving, the chief detects the silent peer
