@@ -132,6 +132,11 @@ class Trainer:
 
     # -- checkpoint state ------------------------------------------------
     def state_tensors(self, include_optimizer: bool = True) -> Dict[str, torch.Tensor]:
+        if self.mode == "async" and getattr(self.engine, "_attached", False):
+            # snapshot the CURRENT shard state, not the chief's last
+            # pull (one apply-round stale); also refreshes global_step
+            self.engine.pull()
+            self.global_step = self.engine.last_global_step
         out: Dict[str, torch.Tensor] = {
             name: p.detach() for name, p in self.model.named_parameters()}
         for name, b in self.model.named_buffers():
