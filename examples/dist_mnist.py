@@ -108,7 +108,10 @@ def main(_):
                 os._exit(17)  # simulated crash (no cleanup, like a real one)
 
     if FLAGS.loss_log:
-        with open(FLAGS.loss_log, "w") as f:
+        # "{task}" lets a shared launcher argument fan out to one log
+        # per worker (rocps-launch passes identical args to every task)
+        with open(FLAGS.loss_log.replace("{task}", str(FLAGS.task_index)),
+                  "w") as f:
             json.dump(losses, f)
     # every worker reports done; the chief shuts the cluster down only
     # once all workers finished (PS must keep serving until then).

@@ -20,19 +20,17 @@ def _launchpy(*argv, timeout):
 
 
 def test_launch_sync_cluster_end_to_end(tmp_path):
-    log0 = str(tmp_path / "w0.json")
+    log_tpl = str(tmp_path / "w{task}.json")  # fans out per worker
     r = _launchpy(
         "--num-ps", "1", "--num-workers", "2", "--num-gpus", "0", "--",
         "examples/dist_mnist.py", "--issync=1", "--train_steps=8",
-        "--batch_size=8", f"--loss_log={log0}", timeout=240)
+        "--batch_size=8", f"--loss_log={log_tpl}", timeout=240)
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "[rocps-launch] ps:0" in r.stdout
     assert "[rocps-launch] worker:1" in r.stdout
-    # worker 0 (chief) wrote its loss trajectory and trained to step 8
-    # (only one --loss_log is given, so both workers get it; last writer
-    # wins — either is a full 8-step log)
-    losses = json.load(open(log0))
-    assert len(losses) >= 8
+    for w in (0, 1):
+        losses = json.load(open(str(tmp_path / f"w{w}.json")))
+        assert len(losses) >= 8
 
 
 def test_launch_propagates_task_failure(tmp_path):
