@@ -114,6 +114,41 @@ def test_keep_checkpoint_every_n_hours(tmp_path):
     assert latest_checkpoint(ckdir).endswith("model.ckpt-3")
 
 
+def test_summary_and_logging_hooks(tmp_path, capsys):
+    import json as _json
+
+    from rocps.obs import MetricsLogger
+    from rocps.train.hooks import LoggingTensorHook, SummarySaverHook
+
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    metrics = MetricsLogger(str(tmp_path))
+    _run(tr, 6, hooks=[SummarySaverHook(metrics, every_n_steps=2),
+                       LoggingTensorHook(every_n_iter=3)])
+    metrics.close()
+    recs = [_json.loads(ln) for ln in
+            (tmp_path / "metrics.jsonl").read_text().strip().split("\n")]
+    assert [r["step"] for r in recs] == [2, 4, 6]
+    assert all("loss" in r for r in recs)
+    out = capsys.readouterr().out
+    assert "[rocps] step 3 loss" in out and "[rocps] step 6 loss" in out
+
+
+def test_stop_at_step_num_steps_relative():
+    """num_steps counts from the session-start step (restore-aware)."""
+    import pytest as _pytest
+
+    from rocps.train.hooks import StopAtStepHook
+
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    tr.global_step = 10
+    _run(tr, 100, hooks=[StopAtStepHook(num_steps=5)])
+    assert tr.global_step == 15
+    with _pytest.raises(ValueError):
+        StopAtStepHook()
+    with _pytest.raises(ValueError):
+        StopAtStepHook(num_steps=1, last_step=2)
+
+
 def test_step_counter_hook():
     tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
     sc = StepCounterHook(every_n_steps=2, batch_size=16)
