@@ -38,6 +38,7 @@ flags.DEFINE_integer("replicas_to_aggregate", 0,
 flags.DEFINE_integer("train_steps", 50, "number of global steps")
 flags.DEFINE_integer("batch_size", 32, "per-worker batch size")
 flags.DEFINE_float("learning_rate", 0.1, "SGD learning rate")
+flags.DEFINE_float("momentum", 0.0, "momentum (0 = plain SGD)")
 flags.DEFINE_string("checkpoint_dir", "", "chief checkpoint dir (optional)")
 flags.DEFINE_integer("save_checkpoint_steps", 0, "ckpt every N steps (0=off)")
 flags.DEFINE_integer("hidden", 32, "MLP hidden width")
@@ -69,7 +70,10 @@ def main(_):
     is_chief = FLAGS.task_index == 0
     torch.manual_seed(FLAGS.seed)
     model = mnist_mlp(hidden=FLAGS.hidden)
-    opt = rocps.train.GradientDescentOptimizer(FLAGS.learning_rate)
+    opt = (rocps.train.MomentumOptimizer(FLAGS.learning_rate,
+                                         FLAGS.momentum)
+           if FLAGS.momentum
+           else rocps.train.GradientDescentOptimizer(FLAGS.learning_rate))
     if FLAGS.issync:
         opt = rocps.train.SyncReplicasOptimizer(
             opt,

@@ -117,7 +117,14 @@ class NanTensorHook(SessionRunHook):
 
 
 class CheckpointSaverHook(SessionRunHook):
-    """Chief-only periodic checkpointing (reference: CheckpointSaverHook)."""
+    """Periodic checkpointing (reference: CheckpointSaverHook).
+
+    Installed on EVERY rank in sync mode: collecting the state is a
+    collective in the reduce-scatter engine (each rank all-gathers its
+    optimizer-state shard), so all ranks must reach it together; only
+    the chief writes the file. Time-based saves are decided by the
+    chief and broadcast so ranks can't straddle the deadline.
+    """
 
     def __init__(self, checkpoint_dir: str, save_secs: Optional[float] = None,
                  save_steps: Optional[int] = None, saver=None,
@@ -131,6 +138,7 @@ class CheckpointSaverHook(SessionRunHook):
         self._basename = checkpoint_basename
         self._last_save_t = time.time()
         self._last_save_s = 0
+        self._is_chief = True
 
     def _prefix(self) -> str:
         import os
@@ -140,20 +148,42 @@ class CheckpointSaverHook(SessionRunHook):
     def after_create_session(self, session) -> None:
         self._saver = self._saver or session.saver
         self._last_save_s = session.global_step
+        self._is_chief = getattr(session, "is_chief", True)
+        self._comm = getattr(session.trainer, "comm", None)
+
+    def _due(self, gs: int) -> bool:
+        if self._save_steps is not None:
+            # deterministic across ranks: global_step is synchronized
+            return gs - self._last_save_s >= self._save_steps
+        due = time.time() - self._last_save_t >= self._save_secs
+        comm = self._comm
+        if comm is not None and comm.is_distributed:
+            # chief decides; ranks at the same step must agree or the
+            # collective state collection deadlocks
+            import torch as _torch
+            flag = _torch.tensor(
+                [1 if (due and self._is_chief) else 0], dtype=_torch.int64,
+                device=comm.device if comm.backend == "nccl" else "cpu")
+            comm.broadcast(flag, src=0)
+            due = bool(flag.item())
+        return due
+
+    def _save(self, session, gs: int) -> None:
+        # async mode has no collectives: non-chief ranks skip collection
+        if self._is_chief or session.trainer.mode == "sync":
+            tensors = session.trainer.state_tensors()  # collective in rs_ag
+            if self._is_chief:
+                self._saver.save(tensors, self._prefix(), global_step=gs)
+        self._last_save_t, self._last_save_s = time.time(), gs
 
     def after_run(self, run_context, run_values) -> None:
         gs = run_context.global_step
-        due = ((self._save_steps is not None and gs - self._last_save_s >= self._save_steps)
-               or (self._save_secs is not None
-                   and time.time() - self._last_save_t >= self._save_secs))
-        if due:
-            self._saver.save(run_context.session, self._prefix(), global_step=gs)
-            self._last_save_t, self._last_save_s = time.time(), gs
+        if self._due(gs):
+            self._save(run_context.session, gs)
 
     def end(self, session) -> None:
-        if self._saver is not None:
-            self._saver.save(session, self._prefix(),
-                             global_step=session.global_step)
+        if self._saver is not None and session.global_step != self._last_save_s:
+            self._save(session, session.global_step)
 
 
 class SummarySaverHook(SessionRunHook):

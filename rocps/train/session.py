@@ -149,8 +149,11 @@ class Trainer:
     def _gather_optimizer_state(self) -> Dict[str, torch.Tensor]:
         out: Dict[str, torch.Tensor] = {}
         eng: SyncEngine = self.engine  # type: ignore[assignment]
+        # SyncReplicasOptimizer delegates apply_flat to .base, so the
+        # slot state (momentum/adam buffers) lives on the base optimizer
+        opt = getattr(self.optimizer, "base", self.optimizer)
         for b in eng.buckets:
-            st = self.optimizer._state.get(("b", b.index))
+            st = opt._state.get(("b", b.index))
             if not st:
                 continue
             for k, v in st.items():
@@ -187,7 +190,8 @@ class Trainer:
                 continue
             target = (b.my_shard(b.params_flat, self.comm.rank)
                       if eng.mode == "rs_ag" else b.params_flat)
-            st = self.optimizer.slot_state(("b", b.index), target)
+            st = getattr(self.optimizer, "base",
+                         self.optimizer).slot_state(("b", b.index), target)
             for key in keys:
                 k = key.rsplit("/", 1)[1]
                 if k not in st:
@@ -220,14 +224,17 @@ class MonitoredTrainingSession:
         self._hooks: List[SessionRunHook] = list(hooks or [])
         if self.is_chief:
             self._hooks += list(chief_only_hooks or [])
-            if checkpoint_dir and not any(
-                    isinstance(h, CheckpointSaverHook) for h in self._hooks):
-                if save_checkpoint_steps is not None:
-                    self._hooks.append(CheckpointSaverHook(
-                        checkpoint_dir, save_steps=save_checkpoint_steps))
-                elif save_checkpoint_secs is not None:
-                    self._hooks.append(CheckpointSaverHook(
-                        checkpoint_dir, save_secs=save_checkpoint_secs))
+        # the saver hook runs on EVERY rank (state collection is a
+        # collective in the reduce-scatter engine; only the chief
+        # writes) — see CheckpointSaverHook
+        if checkpoint_dir and not any(
+                isinstance(h, CheckpointSaverHook) for h in self._hooks):
+            if save_checkpoint_steps is not None:
+                self._hooks.append(CheckpointSaverHook(
+                    checkpoint_dir, save_steps=save_checkpoint_steps))
+            elif save_checkpoint_secs is not None:
+                self._hooks.append(CheckpointSaverHook(
+                    checkpoint_dir, save_secs=save_checkpoint_secs))
 
     # -- properties hooks use -------------------------------------------
     @property
@@ -237,18 +244,25 @@ class MonitoredTrainingSession:
     # -- lifecycle -------------------------------------------------------
     def __enter__(self) -> "MonitoredTrainingSession":
         from rocps.train.saver import latest_checkpoint
-        if self.is_chief and self.checkpoint_dir:
+        if self.checkpoint_dir:
+            # EVERY rank restores (not just the chief): in the reference
+            # the optimizer's slot variables live on the PS, so a chief
+            # restore covers all workers; here sync-mode optimizer state
+            # is sharded/replicated per rank and each rank must load its
+            # own piece — params/global_step are (re)broadcast below.
             ckpt = latest_checkpoint(self.checkpoint_dir)
             if ckpt:
                 self.saver.restore(self.trainer, ckpt)
-                # async mode: the restored weights must overwrite an
-                # already-initialized PS shard (a plain chief restart
-                # without a checkpoint instead ADOPTS the PS state —
-                # reference SessionManager.prepare_session semantics);
-                # the restored global_step continues on the PS counter
-                self.trainer.engine.force_init_load = True
-                self.trainer.engine.init_global_step = \
-                    self.trainer.global_step
+                if self.is_chief:
+                    # async mode: the restored weights must overwrite an
+                    # already-initialized PS shard (a plain chief restart
+                    # without a checkpoint instead ADOPTS the PS state —
+                    # reference SessionManager.prepare_session
+                    # semantics); the restored global_step continues on
+                    # the PS counter
+                    self.trainer.engine.force_init_load = True
+                    self.trainer.engine.init_global_step = \
+                        self.trainer.global_step
         self.trainer.sync_initial_state()
         for h in self._hooks:
             h.begin()

@@ -145,6 +145,31 @@ def test_async_two_workers_two_ps(free_ports, tmp_path):
     assert total_steps >= 40  # cluster-wide step target reached
 
 
+def test_sync_resume_exact_with_momentum(free_ports, tmp_path):
+    """Checkpoint/restore is exact across a cluster restart in sync
+    mode WITH optimizer state: a 12-step run + a resumed 12-step run
+    reproduces the uninterrupted 24-step loss trajectory. Momentum
+    makes this sensitive to every rank restoring its slot buffers
+    (not just the chief)."""
+    mom = ["--momentum=0.9", "--learning_rate=0.05"]
+    straight, _ = _run_cluster(free_ports, tmp_path / "s", issync=True,
+                               steps=24, batch=8, extra=mom)
+    ckdir = str(tmp_path / "ck")
+    _run_cluster(free_ports, tmp_path / "p1", issync=True, steps=12,
+                 batch=8, extra=[*mom, f"--checkpoint_dir={ckdir}",
+                                 "--save_checkpoint_steps=12"])
+    resumed, _ = _run_cluster(free_ports, tmp_path / "p2", issync=True,
+                              steps=24, batch=8,
+                              extra=[*mom, f"--checkpoint_dir={ckdir}",
+                                     "--save_checkpoint_steps=12"])
+    # phase 2 trained only steps 13..24
+    assert len(resumed[0]) == 12 and len(resumed[1]) == 12
+    for s in range(12):
+        a = 0.5 * (straight[0][12 + s] + straight[1][12 + s])
+        b = 0.5 * (resumed[0][s] + resumed[1][s])
+        assert abs(a - b) < 1e-4, (s, a, b)
+
+
 def test_async_checkpoint_resume(free_ports, tmp_path):
     """Async-mode resume across cluster restarts: the chief restores the
     checkpoint, force-loads it into the (fresh) PS shard together with
