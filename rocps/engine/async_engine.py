@@ -209,6 +209,10 @@ class AsyncPSShard:
         # optimizer from manifest
         opt_spec = head.get("optimizer", {"type": "sgd", "lr": 0.01})
         self.optimizer = _build_optimizer(opt_spec)
+        if hasattr(self.optimizer, "_t"):
+            # resumed run: Adam's bias-correction counter continues from
+            # the restored step, like the reference's beta-power slots
+            self.optimizer._t = int(head.get("global_step") or 0)
         self.staleness_bound = head.get("staleness_bound")
         self.replicas_to_aggregate = head.get("replicas_to_aggregate")
         self.flush_timeout = float(head.get("flush_timeout") or 2.0)
@@ -272,6 +276,8 @@ class AsyncPSShard:
                                 and version - seq > self.staleness_bound):
                             self.dropped_stale += 1
                         else:
+                            if hasattr(self.optimizer, "begin_step"):
+                                self.optimizer.begin_step()
                             self.optimizer.apply_flat(
                                 seg.params, seg.grads[w], slot="shard")
                             self.applied += 1
@@ -304,6 +310,8 @@ class AsyncPSShard:
         def apply_round(n):
             nonlocal n_contrib, pending, last_contrib_t
             accum.div_(n)
+            if hasattr(self.optimizer, "begin_step"):
+                self.optimizer.begin_step()
             self.optimizer.apply_flat(seg.params, accum.to(seg.params.dtype),
                                       slot="shard")
             self.applied += 1

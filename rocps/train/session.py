@@ -152,6 +152,9 @@ class Trainer:
         # SyncReplicasOptimizer delegates apply_flat to .base, so the
         # slot state (momentum/adam buffers) lives on the base optimizer
         opt = getattr(self.optimizer, "base", self.optimizer)
+        if hasattr(opt, "_t"):  # Adam bias-correction step counter
+            out["optimizer/t"] = torch.tensor(int(opt._t),
+                                              dtype=torch.int64)
         for b in eng.buckets:
             st = opt._state.get(("b", b.index))
             if not st:
@@ -183,6 +186,9 @@ class Trainer:
 
     def _load_optimizer_state(self, tensors: Dict[str, torch.Tensor]) -> None:
         eng: SyncEngine = self.engine  # type: ignore[assignment]
+        base = getattr(self.optimizer, "base", self.optimizer)
+        if "optimizer/t" in tensors and hasattr(base, "_t"):
+            base._t = int(tensors["optimizer/t"].item())
         for b in eng.buckets:
             keys = [k for k in tensors
                     if k.startswith(f"optimizer/bucket{b.index}/")]

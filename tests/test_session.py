@@ -72,6 +72,34 @@ def test_checkpoint_resume_exact_continuation(tmp_path):
         assert torch.allclose(pA[k], pC[k], atol=1e-7), k
 
 
+def test_checkpoint_resume_exact_adam(tmp_path):
+    """Adam resume must also restore the bias-correction step counter
+    (t): with t reset, the resumed alpha_t is ~2x too large and the
+    trajectories diverge."""
+    from rocps.train import AdamOptimizer
+
+    ckdir = str(tmp_path / "cka")
+    torch.manual_seed(11)
+    mA = mnist_mlp(hidden=16)
+    trA = Trainer(mA, AdamOptimizer(0.01))
+    _run(trA, 30)
+
+    torch.manual_seed(11)
+    mB = mnist_mlp(hidden=16)
+    trB = Trainer(mB, AdamOptimizer(0.01))
+    _run(trB, 15, ckdir=ckdir, save_every=15)
+
+    torch.manual_seed(99)
+    mC = mnist_mlp(hidden=16)
+    trC = Trainer(mC, AdamOptimizer(0.01))
+    _run(trC, 30, ckdir=ckdir, save_every=15)
+    assert trC.global_step == 30
+    pA = dict(mA.named_parameters())
+    pC = dict(mC.named_parameters())
+    for k in pA:
+        assert torch.allclose(pA[k], pC[k], atol=1e-6), k
+
+
 def test_checkpoint_includes_optimizer_state(tmp_path):
     ckdir = str(tmp_path / "ck2")
     tr = Trainer(mnist_mlp(hidden=8), AdamOptimizer(1e-3))
