@@ -33,6 +33,22 @@ def test_launch_sync_cluster_end_to_end(tmp_path):
         assert len(losses) >= 8
 
 
+def test_launch_dist_train_zoo_script(tmp_path):
+    """examples/dist_train.py (the model-zoo variant of the reference
+    script) end-to-end on CPU: mnist_mlp, sync, 2 workers, with JSONL
+    metrics + TB event files in --logdir."""
+    logdir = str(tmp_path / "logs")
+    r = _launchpy(
+        "--num-ps", "1", "--num-workers", "2", "--num-gpus", "0", "--",
+        "examples/dist_train.py", "--model=mnist_mlp", "--issync=1",
+        "--train_steps=6", "--batch_size=8", "--dtype=fp32",
+        f"--logdir={logdir}", timeout=240)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    files = os.listdir(logdir)
+    assert "metrics.jsonl" in files
+    assert any(f.startswith("events.out.tfevents.") for f in files)
+
+
 def test_launch_propagates_task_failure(tmp_path):
     bad = tmp_path / "bad_task.py"
     bad.write_text(
