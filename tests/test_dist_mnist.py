@@ -208,14 +208,18 @@ def test_async_worker_death_tolerated(free_ports, tmp_path):
         ports = free_ports(3)
         ps = [f"127.0.0.1:{ports[0]}"]
         workers = [f"127.0.0.1:{ports[1]}", f"127.0.0.1:{ports[2]}"]
+        # the victim launches FIRST and the step target is large: if the
+        # survivor reached the global step target before the victim's
+        # (slower) startup did 3 local steps, the victim would stop
+        # cleanly (rc 0) instead of crashing
         procs = {
+            "w1": _launch("worker", 1, ps, workers,
+                          ["--issync=0", "--train_steps=300",
+                           "--batch_size=8", "--kill_after_step=3"], d),
             "ps0": _launch("ps", 0, ps, workers, ["--issync=0"], d),
             "w0": _launch("worker", 0, ps, workers,
-                          ["--issync=0", "--train_steps=40",
+                          ["--issync=0", "--train_steps=300",
                            "--batch_size=8", f"--loss_log={d}/l0.json"], d),
-            "w1": _launch("worker", 1, ps, workers,
-                          ["--issync=0", "--train_steps=40",
-                           "--batch_size=8", "--kill_after_step=3"], d),
         }
         try:
             outs = _wait_all(procs, timeout=120)
