@@ -147,6 +147,8 @@ class AsyncPSShard:
         self._apply_thread: Optional[threading.Thread] = None
         server.register_handler("init_shard", self._on_init_shard)
         server.register_handler("shard_stats", self._on_stats)
+        server.register_handler("opt_state_get", self._on_opt_state_get)
+        server.register_handler("opt_state_put", self._on_opt_state_put)
         self.applied = 0
         self.dropped_stale = 0
         self.staleness_bound: Optional[int] = None
@@ -243,6 +245,40 @@ class AsyncPSShard:
                 os.unlink(seg.path)
             except OSError:
                 pass
+
+    def _on_opt_state_get(self, head, payload):
+        """Chief checkpoint path: the reference saves the optimizer's
+        slot variables living on the PS; here that is the shard-slot
+        state (momentum / adam m,v / fp32 master), returned as fp32
+        blobs. Reads race the apply thread by design (same staleness
+        as the chief's parameter pull)."""
+        if self.optimizer is None:
+            return {"ok": False, "error": "shard not initialized"}, b""
+        st = self.optimizer._state.get("shard") or {}
+        keys = sorted(k for k, v in st.items() if torch.is_tensor(v))
+        blobs = [st[k].detach().to(torch.float32).contiguous()
+                 .numpy().tobytes() for k in keys]
+        return ({"ok": True, "keys": keys, "lens": [len(b) for b in blobs],
+                 "t": int(getattr(self.optimizer, "_t", 0))},
+                b"".join(blobs))
+
+    def _on_opt_state_put(self, head, payload):
+        """Checkpoint-restore path: chief pushes the saved slot state
+        back (after a force init_shard re-seeded the params)."""
+        if self.optimizer is None or self.seg is None:
+            return {"ok": False, "error": "shard not initialized"}, b""
+        st = self.optimizer.slot_state("shard", self.seg.params)
+        off = 0
+        for k, ln in zip(head.get("keys", []), head.get("lens", [])):
+            arr = np.frombuffer(payload, dtype=np.float32, count=ln // 4,
+                                offset=off)
+            if k in st and torch.is_tensor(st[k]):
+                st[k].copy_(torch.from_numpy(arr.copy())
+                            .reshape(st[k].shape))
+            off += ln
+        if hasattr(self.optimizer, "_t") and head.get("t") is not None:
+            self.optimizer._t = int(head["t"])
+        return {"ok": True}, b""
 
     def _on_stats(self, head, payload):
         return {"ok": True, "applied": self.applied,
@@ -490,8 +526,35 @@ class AsyncWorkerEngine:
                         time.sleep(0.2)
                         continue
                     raise RuntimeError(f"init_shard failed: {r}")
+                restored = getattr(self, "restored_opt_state", None)
+                if force and restored and s in restored:
+                    keys, blobs, t = restored[s]
+                    r = c.call("opt_state_put", payload=b"".join(blobs),
+                               keys=keys, lens=[len(b) for b in blobs],
+                               t=t)
+                    if not r.get("ok"):
+                        raise RuntimeError(f"opt_state_put failed: {r}")
                 c.close()
         self._attach()
+
+    def fetch_opt_state(self, shard: int):
+        """(keys, fp32 tensors, t) of the PS shard's optimizer slots."""
+        ps_addr = self.cluster.task_address(
+            "ps", self.cluster.task_indices("ps")[shard])
+        c = ControlClient(ps_addr)
+        try:
+            r, payload = c.call_raw("opt_state_get")
+        finally:
+            c.close()
+        if not r.get("ok"):
+            return [], [], 0
+        out, off = [], 0
+        for ln in r["lens"]:
+            arr = np.frombuffer(payload, dtype=np.float32, count=ln // 4,
+                                offset=off)
+            out.append(torch.from_numpy(arr.copy()))
+            off += ln
+        return list(r["keys"]), out, int(r.get("t", 0))
 
     def _attach(self) -> None:
         for s in range(self.num_shards):

@@ -144,6 +144,16 @@ class Trainer:
         out["global_step"] = torch.tensor(self.global_step, dtype=torch.int64)
         if include_optimizer and self.mode == "sync":
             out.update(self._gather_optimizer_state())
+        elif include_optimizer and self.mode == "async" and getattr(
+                self.engine, "_attached", False):
+            # the reference checkpoints the optimizer slot variables
+            # living on the PS; fetch each shard's slot state
+            for s in range(self.engine.num_shards):
+                keys, tensors, t = self.engine.fetch_opt_state(s)
+                for k, v in zip(keys, tensors):
+                    out[f"optimizer/shard{s}/{k}"] = v
+                if t:
+                    out["optimizer/t"] = torch.tensor(t, dtype=torch.int64)
         return out
 
     def _gather_optimizer_state(self) -> Dict[str, torch.Tensor]:
@@ -183,6 +193,22 @@ class Trainer:
                     bufs[name].copy_(t.to(bufs[name].device))
         if self.mode == "sync":
             self._load_optimizer_state(tensors)
+        elif self.mode == "async":
+            # stash PS slot state; the engine pushes it after the force
+            # init_shard re-seeds the shard params (chief restore path)
+            restored: Dict[int, tuple] = {}
+            t = int(tensors["optimizer/t"].item()) \
+                if "optimizer/t" in tensors else self.global_step
+            for name, v in tensors.items():
+                if not name.startswith("optimizer/shard"):
+                    continue
+                sh, key = name[len("optimizer/shard"):].split("/", 1)
+                keys, blobs, _ = restored.setdefault(int(sh), ([], [], t))
+                keys.append(key)
+                blobs.append(v.to(torch.float32).contiguous()
+                             .numpy().tobytes())
+            if restored:
+                self.engine.restored_opt_state = restored
 
     def _load_optimizer_state(self, tensors: Dict[str, torch.Tensor]) -> None:
         eng: SyncEngine = self.engine  # type: ignore[assignment]

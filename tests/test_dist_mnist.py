@@ -178,15 +178,20 @@ def test_async_checkpoint_resume(free_ports, tmp_path):
     from rocps.train.saver import latest_checkpoint
 
     ckdir = str(tmp_path / "ack")
+    mom = ["--momentum=0.9", f"--checkpoint_dir={ckdir}",
+           "--save_checkpoint_steps=4"]
     p1, _ = _run_cluster(
         free_ports, tmp_path / "p1", issync=False, n_workers=1, n_ps=1,
-        steps=8, batch=16,
-        extra=[f"--checkpoint_dir={ckdir}", "--save_checkpoint_steps=4"])
+        steps=8, batch=16, extra=mom)
     assert latest_checkpoint(ckdir).endswith("model.ckpt-8")
+    # the checkpoint carries the PS-side optimizer slot state
+    from rocps.train.saver import NewCheckpointReader
+    r = NewCheckpointReader(latest_checkpoint(ckdir))
+    assert any(k.startswith("optimizer/shard0/")
+               for k in r.get_variable_to_shape_map())
     p2, _ = _run_cluster(
         free_ports, tmp_path / "p2", issync=False, n_workers=1, n_ps=1,
-        steps=14, batch=16,
-        extra=[f"--checkpoint_dir={ckdir}", "--save_checkpoint_steps=4"])
+        steps=14, batch=16, extra=mom)
     # resumed at step 8: only ~6 further steps, counter continued to 14
     assert latest_checkpoint(ckdir).endswith("model.ckpt-14")
     assert len(p2[0]) <= 8  # not re-trained from scratch
@@ -333,6 +338,73 @@ def test_async_worker_restart_reattaches():
         loss3 = float(m3.loss(x, y).detach())
         assert abs(loss3 - first[0]) < 1e-5, (loss3, first[0])
         eng3.detach()
+    finally:
+        for s in workers + ps:
+            s.stop()
+
+
+def test_async_ps_opt_state_roundtrip():
+    """C12 parity for async mode: the PS shard's optimizer slot state
+    (momentum buffer here) is fetchable for checkpoints and restorable
+    into a fresh cluster, continuing the optimizer trajectory."""
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import MomentumOptimizer
+
+    g = torch.Generator().manual_seed(1)
+    x, y = synthetic_mnist_batch(64, generator=g)
+
+    def start_cluster():
+        workers, ps, cluster = create_local_cluster(num_workers=1, num_ps=1)
+        shard = AsyncPSShard.serve_on(ps[0], cluster)
+        return workers, ps, cluster, shard
+
+    workers, ps, cluster, shard = start_cluster()
+    try:
+        torch.manual_seed(5)
+        m = mnist_mlp(hidden=16)
+        eng = AsyncWorkerEngine(m, cluster, ControlClient(cluster.master_address),
+                                task_index=0, device=torch.device("cpu"))
+        opt = MomentumOptimizer(0.2, 0.9)
+        eng.initialize(opt)
+        for _ in range(5):
+            eng.zero_grad()
+            eng.step(m.loss(x, y), opt)
+        keys, blobs, t = eng.fetch_opt_state(0)
+        assert "mom" in keys
+        mom = blobs[keys.index("mom")]
+        assert mom.abs().sum() > 0  # momentum accumulated
+        params_after = eng.flat_params[0].clone()
+        eng.detach()
+    finally:
+        for s in workers + ps:
+            s.stop()
+
+    # "restart": fresh cluster; push params + slot state back
+    workers, ps, cluster, shard = start_cluster()
+    try:
+        torch.manual_seed(5)
+        m2 = mnist_mlp(hidden=16)
+        eng2 = AsyncWorkerEngine(m2, cluster,
+                                 ControlClient(cluster.master_address),
+                                 task_index=0, device=torch.device("cpu"))
+        with torch.no_grad():
+            eng2.flat_params[0].copy_(params_after)  # "restored" weights
+        eng2.restored_opt_state = {0: (keys, [b.numpy().tobytes()
+                                             for b in blobs], 5)}
+        eng2.initialize(MomentumOptimizer(0.2, 0.9), force_load=True)
+        k2, b2, _ = eng2.fetch_opt_state(0)
+        assert torch.equal(b2[k2.index("mom")], mom)
+        # one more step uses the restored momentum: the PS update moves
+        # params by mom*momentum + grad, not by grad alone
+        eng2.zero_grad()
+        eng2.step(m2.loss(x, y), MomentumOptimizer(0.2, 0.9))
+        assert shard.applied == 1
+        eng2.detach()
     finally:
         for s in workers + ps:
             s.stop()
