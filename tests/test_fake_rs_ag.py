@@ -92,7 +92,68 @@ def test_fake_four_ranks_momentum_rs_ag():
         assert torch.allclose(p, p_ref[n], atol=1e-5), n
 
 
-def test_fake_collectives_primitives():
+def test_fake_rs_ag_checkpoint_save_restore_sharded_state():
+    """Trainer-level save/restore over the rs_ag engine: collecting the
+    checkpoint all-gathers each rank's optimizer-state shard and restore
+    re-scatters it — the exact path the RCCL engine takes on GPU.
+    12 straight steps == 6 steps + save + restore-into-fresh-world + 6."""
+    from rocps.train.session import Trainer
+
+    steps, B, lr, mu = 6, 8, 0.1, 0.9
+
+    def trainer_run(models, steps, start_step, state=None, out_state=None):
+        results = {}
+        lock = threading.Lock()
+
+        def rank_fn(comm: FakeComm, rank: int):
+            tr = Trainer(models[rank], MomentumOptimizer(lr, mu),
+                         comm=comm, sync_mode="rs_ag", bucket_cap_mb=0.01)
+            tr.global_step = start_step
+            if state is not None:
+                tr.load_state_tensors(state)
+            tr.engine.broadcast_parameters(src=0)
+            for s in range(tr.global_step, tr.global_step + steps):
+                g = torch.Generator().manual_seed(3 + s)
+                x, y = synthetic_mnist_batch(2 * B, generator=g)
+                lo = rank * B
+                xb, yb = x[lo:lo + B], y[lo:lo + B]
+                tr.train_step(lambda: tr.model.loss(xb, yb))
+            st = tr.state_tensors()  # collective on every rank
+            with lock:
+                results[rank] = st
+            return rank
+
+        run_fake_cluster(2, rank_fn)
+        if out_state is not None:
+            out_state.update(results[0])
+        return results
+
+    # straight 12-step reference
+    models = []
+    for _ in range(2):
+        torch.manual_seed(7)
+        models.append(mnist_mlp(hidden=32))
+    straight = trainer_run(models, 12, 0)
+
+    # 6 steps, checkpoint, restore into a FRESH world, 6 more
+    models = []
+    for _ in range(2):
+        torch.manual_seed(7)
+        models.append(mnist_mlp(hidden=32))
+    saved = {}
+    trainer_run(models, steps, 0, out_state=saved)
+    assert saved["global_step"].item() == 6
+    assert any(k.startswith("optimizer/bucket") for k in saved)
+
+    models = []
+    for _ in range(2):
+        torch.manual_seed(99)  # different init; restore must overwrite
+        models.append(mnist_mlp(hidden=32))
+    resumed = trainer_run(models, steps, 6, state=saved)
+
+    for k, v in straight[0].items():
+        assert torch.allclose(v.float(), resumed[0][k].float(),
+                              atol=1e-5), k
     def rank_fn(comm: FakeComm, rank: int):
         t = torch.full((4,), float(rank + 1))
         comm.all_reduce(t)
