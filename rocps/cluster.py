@@ -329,8 +329,11 @@ class Server:
                 daemon=True)
             self._thread.start()
         # Register with the cluster master (self-registration included).
+        # Few retries only: a master that is not up yet must not stall
+        # this task's startup — heartbeats re-register within seconds.
         try:
-            ControlClient(self.cluster.master_address).register(
+            ControlClient(self.cluster.master_address,
+                          connect_retries=8).register(
                 self.job_name, self.task_index)
         except Exception:
             pass  # master may not be up yet; heartbeats will register later

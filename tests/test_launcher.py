@@ -49,6 +49,66 @@ def test_launch_dist_train_zoo_script(tmp_path):
     assert any(f.startswith("events.out.tfevents.") for f in files)
 
 
+def test_ps_server_cli(tmp_path):
+    """rocps.cli.ps_server (the standalone PS task bench.py spawns):
+    serve a shard to an in-process worker engine, then shut down."""
+    import socket
+
+    import torch
+
+    from rocps.cluster import ClusterSpec, ControlClient, Server, \
+        shutdown_cluster
+    from rocps.engine.async_engine import AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    socks, ports = [], []
+    for _ in range(2):
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        socks.append(s)
+        ports.append(s.getsockname()[1])
+    for s in socks:
+        s.close()
+    spec = {"ps": [f"127.0.0.1:{ports[0]}"],
+            "worker": [f"127.0.0.1:{ports[1]}"]}
+    cluster = ClusterSpec(spec)
+    # spawn the ps FIRST: it hosts the cluster master (kv), and a
+    # starting Server retries registration against it
+    ps = subprocess.Popen(
+        [sys.executable, "-m", "rocps.cli.ps_server",
+         "--cluster-json", json.dumps(spec), "--task-index", "0"],
+        cwd=REPO, env=dict(os.environ, ROCPS_BIND_HOST="127.0.0.1"),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    worker_server = Server(cluster, "worker", 0)
+    try:
+        torch.manual_seed(5)
+        m = mnist_mlp(hidden=16)
+        eng = AsyncWorkerEngine(m, cluster,
+                                ControlClient(cluster.master_address),
+                                task_index=0, device=torch.device("cpu"))
+        opt = GradientDescentOptimizer(0.5)
+        eng.initialize(opt)
+        g = torch.Generator().manual_seed(1)
+        x, y = synthetic_mnist_batch(32, generator=g)
+        first = last = None
+        for _ in range(8):
+            eng.zero_grad()
+            loss = m.loss(x, y)
+            eng.step(loss, opt)
+            last = float(loss.detach())
+            first = first if first is not None else last
+        assert last < first
+        eng.detach()
+        shutdown_cluster(cluster)
+        out, _ = ps.communicate(timeout=60)
+        assert ps.returncode == 0, out.decode(errors="replace")
+    finally:
+        if ps.poll() is None:
+            ps.kill()
+        worker_server.stop()
+
+
 def test_launch_propagates_task_failure(tmp_path):
     bad = tmp_path / "bad_task.py"
     bad.write_text(
