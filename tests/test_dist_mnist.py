@@ -494,6 +494,29 @@ def _semisync_once():
             s.stop()
 
 
+def test_semisync_checkpoint_resume(free_ports, tmp_path):
+    """Backup-worker mode (K=2 of 3) resumes from a checkpoint: the PS
+    round/step counters and shard optimizer state continue."""
+    from rocps.train.saver import NewCheckpointReader, latest_checkpoint
+
+    ckdir = str(tmp_path / "sck")
+    extra = ["--replicas_to_aggregate=2", "--momentum=0.9",
+             f"--checkpoint_dir={ckdir}", "--save_checkpoint_steps=5"]
+    _run_cluster(free_ports, tmp_path / "p1", issync=True, n_workers=3,
+                 n_ps=1, steps=10, batch=8, extra=extra)
+    ck = latest_checkpoint(ckdir)
+    assert ck is not None
+    step1 = int(NewCheckpointReader(ck).get_tensor("global_step"))
+    assert step1 >= 10
+    assert any(k.startswith("optimizer/shard0/")
+               for k in NewCheckpointReader(ck).get_variable_to_shape_map())
+    _run_cluster(free_ports, tmp_path / "p2", issync=True, n_workers=3,
+                 n_ps=1, steps=step1 + 10, batch=8, extra=extra)
+    ck2 = latest_checkpoint(ckdir)
+    step2 = int(NewCheckpointReader(ck2).get_tensor("global_step"))
+    assert step2 >= step1 + 10  # continued, not restarted
+
+
 def test_semisync_subprocess_cluster(free_ports, tmp_path):
     """--issync with --replicas_to_aggregate=2 of 3 workers end-to-end
     through the reference's CLI surface."""
