@@ -15,7 +15,8 @@ from rocps.train.session import (  # noqa: F401
     MonitoredTrainingSession, Trainer)
 from rocps.train.hooks import (  # noqa: F401
     SessionRunHook, StopAtStepHook, StepCounterHook, NanTensorHook,
-    CheckpointSaverHook, SummarySaverHook, LoggingTensorHook)
+    CheckpointSaverHook, SummarySaverHook, LoggingTensorHook,
+    ProfilerHook)
 from rocps.train.saver import (  # noqa: F401
     Saver, latest_checkpoint, NewCheckpointReader)
 from rocps.train.coordinator import Coordinator, LooseVersionThread  # noqa: F401

@@ -10,11 +10,13 @@ from __future__ import annotations
 
 import math
 import time
+
+import torch
 from typing import Any, Dict, Optional
 
 __all__ = ["SessionRunHook", "StopAtStepHook", "StepCounterHook",
            "NanTensorHook", "CheckpointSaverHook", "SummarySaverHook",
-           "LoggingTensorHook"]
+           "LoggingTensorHook", "ProfilerHook"]
 
 
 class RunContext:
@@ -184,6 +186,51 @@ class CheckpointSaverHook(SessionRunHook):
     def end(self, session) -> None:
         if self._saver is not None and session.global_step != self._last_save_s:
             self._save(session, session.global_step)
+
+
+class ProfilerHook(SessionRunHook):
+    """Chrome-trace capture every N steps (reference: tf.train.
+    ProfilerHook / RunMetadata timelines). Uses torch.profiler, which on
+    ROCm records HIP kernel + memcpy activity; traces open in
+    chrome://tracing or perfetto."""
+
+    def __init__(self, save_steps: int, output_dir: str,
+                 profile_steps: int = 1):
+        self._save_steps = int(save_steps)
+        self._dir = output_dir
+        self._profile_steps = int(profile_steps)
+        self._prof = None
+        self._steps_left = 0
+
+    def before_run(self, run_context) -> None:
+        gs = run_context.global_step
+        if self._prof is None and gs and gs % self._save_steps == 0:
+            import torch.profiler as tp
+            acts = [tp.ProfilerActivity.CPU]
+            if torch.cuda.is_available():
+                acts.append(tp.ProfilerActivity.CUDA)
+            self._prof = tp.profile(activities=acts)
+            self._prof.__enter__()
+            self._steps_left = self._profile_steps
+            self._start_step = gs
+
+    def after_run(self, run_context, run_values) -> None:
+        if self._prof is None:
+            return
+        self._steps_left -= 1
+        if self._steps_left <= 0:
+            import os
+            self._prof.__exit__(None, None, None)
+            os.makedirs(self._dir, exist_ok=True)
+            path = os.path.join(self._dir,
+                                f"trace_step{self._start_step}.json")
+            self._prof.export_chrome_trace(path)
+            self._prof = None
+
+    def end(self, session) -> None:
+        if self._prof is not None:
+            self._prof.__exit__(None, None, None)
+            self._prof = None
 
 
 class SummarySaverHook(SessionRunHook):

@@ -177,6 +177,20 @@ def test_stop_at_step_num_steps_relative():
         StopAtStepHook(num_steps=1, last_step=2)
 
 
+def test_profiler_hook(tmp_path):
+    import json as _json
+
+    from rocps.train.hooks import ProfilerHook
+
+    tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
+    pdir = str(tmp_path / "prof")
+    _run(tr, 7, hooks=[ProfilerHook(save_steps=3, output_dir=pdir)])
+    files = sorted(os.listdir(pdir))
+    assert "trace_step3.json" in files and "trace_step6.json" in files
+    doc = _json.load(open(os.path.join(pdir, "trace_step3.json")))
+    assert doc["traceEvents"]  # valid chrome trace with events
+
+
 def test_step_counter_hook():
     tr = Trainer(mnist_mlp(hidden=8), GradientDescentOptimizer(0.1))
     sc = StepCounterHook(every_n_steps=2, batch_size=16)
