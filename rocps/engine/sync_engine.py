@@ -68,9 +68,24 @@ class SyncEngine:
 
     # ------------------------------------------------------------------
     def broadcast_parameters(self, src: int = 0) -> None:
-        """Chief init broadcast (reference: chief runs init_op, others wait)."""
+        """Chief init broadcast (reference: chief runs init_op, others wait).
+        Includes non-parameter buffers (BN running stats): the reference
+        keeps them as shared PS variables, so replicas start identical."""
         for b in self.buckets:
             self.comm.broadcast(b.params_flat, src=src)
+        for _, buf in self.model.named_buffers():
+            self.comm.broadcast(buf, src=src)
+
+    def sync_buffers(self) -> None:
+        """All-reduce-average float buffers (BN running stats) so every
+        replica agrees — the eager recast of the reference's PS-shared
+        moving statistics. Collective: call on every rank together."""
+        if not self.comm.is_distributed:
+            return
+        for _, buf in self.model.named_buffers():
+            if buf.is_floating_point():
+                self.comm.all_reduce(buf)
+                buf.div_(self.comm.world_size)
 
     def zero_grad(self) -> None:
         for b in self.buckets:

@@ -137,6 +137,11 @@ class Trainer:
             # pull (one apply-round stale); also refreshes global_step
             self.engine.pull()
             self.global_step = self.engine.last_global_step
+        if self.mode == "sync" and hasattr(self.engine, "sync_buffers"):
+            # BN running stats are rank-local between saves; average
+            # them across replicas (the reference's PS-shared moving
+            # stats), so the checkpoint sees cluster-wide statistics
+            self.engine.sync_buffers()
         out: Dict[str, torch.Tensor] = {
             name: p.detach() for name, p in self.model.named_parameters()}
         for name, b in self.model.named_buffers():

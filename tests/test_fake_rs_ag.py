@@ -92,6 +92,52 @@ def test_fake_four_ranks_momentum_rs_ag():
         assert torch.allclose(p, p_ref[n], atol=1e-5), n
 
 
+def test_fake_bn_buffer_sync():
+    """BN running stats are rank-local between saves; state collection
+    averages them across replicas (the reference keeps moving stats as
+    shared PS variables)."""
+    from rocps.train.session import Trainer
+
+    class TinyBN(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(4, 4)
+            self.bn = torch.nn.BatchNorm1d(4)
+
+        def loss(self, x, y):
+            return ((self.bn(self.fc(x)) - y) ** 2).mean()
+
+    models = []
+    for _ in range(2):
+        torch.manual_seed(7)
+        models.append(TinyBN())
+    results = {}
+    lock = threading.Lock()
+
+    def rank_fn(comm: FakeComm, rank: int):
+        tr = Trainer(models[rank], GradientDescentOptimizer(0.01),
+                     comm=comm, sync_mode="rs_ag", bucket_cap_mb=0.01)
+        g = torch.Generator().manual_seed(100 + rank)  # DIFFERENT data
+        x = torch.randn(8, 4, generator=g) * (1.0 + rank)
+        y = torch.zeros(8, 4)
+        for _ in range(3):
+            tr.train_step(lambda: tr.model.loss(x, y))
+        local_mean = tr.model.bn.running_mean.clone()
+        st = tr.state_tensors()  # collective: averages buffers
+        with lock:
+            results[rank] = (local_mean, st,
+                             tr.model.bn.running_mean.clone())
+        return rank
+
+    run_fake_cluster(2, rank_fn)
+    m0, st0, after0 = results[0]
+    m1, st1, after1 = results[1]
+    assert not torch.allclose(m0, m1)  # drifted apart (different data)
+    expect = 0.5 * (m0 + m1)
+    assert torch.allclose(st0["bn.running_mean"], expect, atol=1e-6)
+    assert torch.equal(after0, after1)  # replicas re-synced in place
+
+
 def test_fake_rs_ag_checkpoint_save_restore_sharded_state():
     """Trainer-level save/restore over the rs_ag engine: collecting the
     checkpoint all-gathers each rank's optimizer-state shard and restore
