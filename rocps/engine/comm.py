@@ -54,8 +54,19 @@ class Comm:
     # ------------------------------------------------------------------
     @staticmethod
     def tune_env() -> None:
-        """RCCL tuning for 7-link xGMI (call before init_process_group)."""
-        os.environ.setdefault("NCCL_MIN_NCHANNELS", "8")
+        """RCCL tuning for 7-link xGMI (call before init_process_group).
+
+        ROCPS_DETERMINISTIC=1 (SURVEY §5 debug mode) instead pins RCCL
+        to one channel and a fixed ring/simple schedule so the
+        floating-point reduction order is identical run-to-run; the
+        engine also serializes bucket launches under the same flag."""
+        if os.environ.get("ROCPS_DETERMINISTIC") == "1":
+            os.environ.setdefault("NCCL_MIN_NCHANNELS", "1")
+            os.environ.setdefault("NCCL_MAX_NCHANNELS", "1")
+            os.environ.setdefault("NCCL_ALGO", "Ring")
+            os.environ.setdefault("NCCL_PROTO", "Simple")
+        else:
+            os.environ.setdefault("NCCL_MIN_NCHANNELS", "8")
         os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 
     @classmethod

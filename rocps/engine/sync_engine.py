@@ -26,6 +26,7 @@ barrier), chosen for 7-link xGMI instead of a PS hop:
 
 from __future__ import annotations
 
+import os
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -42,6 +43,13 @@ class SyncEngine:
                  bucket_cap_mb: float = 25.0,
                  grad_dtype: Optional[torch.dtype] = None,
                  overlap: bool = True):
+        # ROCPS_DETERMINISTIC=1 (SURVEY §5 race-debug mode): disable
+        # hook-order collective launches — every bucket launches in
+        # bucket-index order at finish_backward, giving a fixed
+        # reduction schedule run-to-run. Comm.tune_env pins RCCL to a
+        # single channel/fixed algorithm under the same flag.
+        if os.environ.get("ROCPS_DETERMINISTIC") == "1":
+            overlap = False
         self.comm = comm
         self.model = model
         self.mode = mode or ("rs_ag" if comm.supports_reduce_scatter()

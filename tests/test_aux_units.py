@@ -245,11 +245,33 @@ def test_comm_world1_shortcircuits():
 def test_comm_tune_env(monkeypatch):
     from rocps.engine.comm import Comm
     monkeypatch.delenv("NCCL_MIN_NCHANNELS", raising=False)
+    monkeypatch.delenv("ROCPS_DETERMINISTIC", raising=False)
     monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     Comm.tune_env()
     # >=8 channels so rings stripe across the 7 xGMI links
     assert int(os.environ["NCCL_MIN_NCHANNELS"]) >= 8
     assert os.environ["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
+
+
+def test_deterministic_mode(monkeypatch):
+    """ROCPS_DETERMINISTIC=1: one RCCL channel + fixed schedule, and
+    the sync engine serializes bucket launches (overlap off)."""
+    from rocps.engine.comm import Comm
+    from rocps.engine.sync_engine import SyncEngine
+    from rocps.models import mnist_mlp
+
+    monkeypatch.setenv("ROCPS_DETERMINISTIC", "1")
+    for k in ("NCCL_MIN_NCHANNELS", "NCCL_MAX_NCHANNELS",
+              "NCCL_ALGO", "NCCL_PROTO"):
+        monkeypatch.delenv(k, raising=False)
+    Comm.tune_env()
+    assert os.environ["NCCL_MAX_NCHANNELS"] == "1"
+    assert os.environ["NCCL_ALGO"] == "Ring"
+    eng = SyncEngine(mnist_mlp(hidden=8),
+                     Comm(device=torch.device("cpu"), backend="gloo"),
+                     mode="allreduce", bucket_cap_mb=0.01)
+    assert eng.overlap is False
+    eng.detach()
 
 
 # ---------------------------------------------------------------------------
