@@ -132,6 +132,10 @@ class Trainer:
 
     # -- checkpoint state ------------------------------------------------
     def state_tensors(self, include_optimizer: bool = True) -> Dict[str, torch.Tensor]:
+        """Checkpoint state snapshot. COLLECTIVE in distributed sync
+        mode (buffer averaging + rs_ag optimizer-state all-gather):
+        every rank must call it at the same point — CheckpointSaverHook
+        does this; don't call it from only one rank."""
         if self.mode == "async" and getattr(self.engine, "_attached", False):
             # snapshot the CURRENT shard state, not the chief's last
             # pull (one apply-round stale); also refreshes global_step
