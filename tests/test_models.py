@@ -1,0 +1,59 @@
+"""Model zoo construction + forward/loss shape checks (SURVEY §2a:
+the reference's model set rebuilt — MNIST MLP, ResNet-50,
+Wide-ResNet-101-2, BERT-base). Heavy models run one tiny CPU forward;
+full training steps are covered by the engine/GPU tests.
+"""
+
+import pytest
+import torch
+
+from rocps.models import build_model
+
+
+def _n_params(m):
+    return sum(p.numel() for p in m.parameters())
+
+
+def test_build_model_unknown():
+    with pytest.raises(ValueError):
+        build_model("alexnet")
+
+
+def test_mnist_mlp_loss():
+    m = build_model("mnist_mlp", hidden=32)
+    x, y = torch.rand(4, 784), torch.randint(0, 10, (4,))
+    loss = m.loss(x, y)
+    assert loss.ndim == 0 and torch.isfinite(loss)
+
+
+def test_resnet50_shape():
+    m = build_model("resnet50")
+    # torchvision-style resnet50: 25.56 M params, 1000-way head
+    assert abs(_n_params(m) - 25_557_032) < 20_000
+    m.eval()
+    with torch.no_grad():
+        out = m(torch.randn(1, 3, 64, 64))
+    assert out.shape == (1, 1000)
+    loss = m.loss(torch.randn(2, 3, 64, 64), torch.randint(0, 1000, (2,)))
+    assert torch.isfinite(loss)
+
+
+def test_wide_resnet101_2_shape():
+    m = build_model("wide_resnet101_2")
+    # 126.89 M params (torchvision wide_resnet101_2)
+    assert abs(_n_params(m) - 126_886_696) < 50_000
+    m.eval()
+    with torch.no_grad():
+        out = m(torch.randn(1, 3, 64, 64))
+    assert out.shape == (1, 1000)
+
+
+def test_bert_base_shape():
+    m = build_model("bert_base")
+    # BERT-base: ~110 M params incl. MLM head tie
+    n = _n_params(m)
+    assert 100e6 < n < 135e6, n
+    ids = torch.randint(0, 30522, (2, 16))
+    lbl = torch.randint(0, 30522, (2, 16))
+    loss = m.loss(ids, lbl)
+    assert torch.isfinite(loss)
