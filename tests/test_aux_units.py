@@ -253,6 +253,17 @@ def test_comm_tune_env(monkeypatch):
     assert os.environ["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
 
 
+def test_comm_from_env_single(monkeypatch):
+    """bench.py single-process path: no WORLD_SIZE -> world-1 comm with
+    no process group."""
+    from rocps.engine.comm import Comm
+    for k in ("WORLD_SIZE", "RANK", "LOCAL_RANK"):
+        monkeypatch.delenv(k, raising=False)
+    c = Comm.from_env()
+    assert c.world_size == 1 and c.rank == 0
+    assert not c.is_distributed
+
+
 def test_deterministic_mode(monkeypatch):
     """ROCPS_DETERMINISTIC=1: one RCCL channel + fixed schedule, and
     the sync engine serializes bucket launches (overlap off)."""
