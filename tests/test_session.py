@@ -125,6 +125,24 @@ def test_max_to_keep_gc(tmp_path):
     assert Saver.last_checkpoints(ckdir) == ["model.ckpt-2", "model.ckpt-3"]
 
 
+def test_saver_raw_module_source(tmp_path):
+    """Saver accepts a bare nn.Module (state_dict save / load_state_dict
+    restore), not just a Trainer."""
+    torch.manual_seed(1)
+    m = mnist_mlp(hidden=8)
+    s = Saver(m)
+    prefix = s.save(None, str(tmp_path / "raw" / "model.ckpt"),
+                    global_step=3)
+    assert prefix.endswith("-3")
+    torch.manual_seed(2)
+    m2 = mnist_mlp(hidden=8)
+    assert not all(torch.equal(a, b) for (_, a), (_, b)
+                   in zip(m.named_parameters(), m2.named_parameters()))
+    Saver(m2).restore(None, prefix)
+    for (n, a), (_, b) in zip(m.named_parameters(), m2.named_parameters()):
+        assert torch.equal(a, b), n
+
+
 def test_keep_checkpoint_every_n_hours(tmp_path):
     """n=0: every GC'd checkpoint is preserved on disk (off the state
     file); the default (10000h) deletes as usual — reference Saver
