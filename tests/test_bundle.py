@@ -185,3 +185,49 @@ def test_empty_bundle(tmp_path):
     prefix = str(tmp_path / "none.ckpt")
     write_bundle(prefix, {})
     assert read_bundle(prefix) == {}
+
+
+# ---------------------------------------------------------------------------
+# property-based fuzz: arbitrary name/shape/dtype sets round-trip
+# ---------------------------------------------------------------------------
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+_DTYPES = [torch.float32, torch.float64, torch.float16, torch.bfloat16,
+           torch.int32, torch.int64, torch.bool]
+_name = st.text(
+    alphabet=st.sampled_from("abcdefgh_/0123456789"), min_size=1,
+    max_size=24).filter(lambda s: not s.startswith("/") and "//" not in s)
+_entry = st.tuples(
+    _name,
+    st.lists(st.integers(0, 7), min_size=0, max_size=3),  # shape, 0-dim ok
+    st.sampled_from(range(len(_DTYPES))),
+)
+
+
+@settings(deadline=None, max_examples=30)
+@given(st.lists(_entry, min_size=0, max_size=12,
+                unique_by=lambda e: e[0]))
+def test_bundle_fuzz_roundtrip(entries):
+    import tempfile
+
+    g = torch.Generator().manual_seed(0)
+    tensors = {}
+    for name, shape, di in entries:
+        dt = _DTYPES[di]
+        if dt.is_floating_point:
+            t = torch.randn(shape, generator=g).to(dt)
+        elif dt == torch.bool:
+            t = torch.rand(shape, generator=g) > 0.5
+        else:
+            t = torch.randint(-1000, 1000, shape, generator=g, dtype=dt)
+        tensors[name] = t
+    with tempfile.TemporaryDirectory() as d:
+        prefix = os.path.join(d, "f.ckpt")
+        write_bundle(prefix, tensors)
+        back = read_bundle(prefix)
+    assert set(back) == set(tensors)
+    for k, t in tensors.items():
+        assert back[k].dtype == t.dtype, k
+        assert back[k].shape == t.shape, k
+        assert torch.equal(back[k], t), k
