@@ -25,6 +25,42 @@ def test_buckets_structure():
             assert p.grad.data_ptr() == b.grads[off:].data_ptr()
 
 
+def test_buckets_fuzz_invariants():
+    """Property fuzz: arbitrary parameter shapes / world sizes keep the
+    bucket invariants — full coverage, world*64 padding, shard split,
+    view aliasing, value preservation through the repack."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(deadline=None, max_examples=25)
+    @given(
+        shapes=st.lists(st.lists(st.integers(1, 9), min_size=1,
+                                 max_size=3), min_size=1, max_size=10),
+        world=st.sampled_from([1, 2, 4, 8]),
+        cap_kb=st.sampled_from([1, 4, 64]),
+    )
+    def check(shapes, world, cap_kb):
+        params = [(f"p{i}", torch.nn.Parameter(torch.randn(s)))
+                  for i, s in enumerate(shapes)]
+        originals = {n: p.detach().clone() for n, p in params}
+        buckets = build_buckets(params, world_size=world,
+                                bucket_cap_mb=cap_kb / 1024,
+                                flatten_params=True)
+        total = sum(v.numel() for v in originals.values())
+        assert sum(b.numel for b in buckets) == total
+        seen = set()
+        for b in buckets:
+            assert b.numel_padded % (world * 64) == 0
+            assert b.shard_numel * world == b.numel_padded
+            for (name, p), off in zip(b.named_params, b.offsets):
+                assert name not in seen
+                seen.add(name)
+                assert p.data.data_ptr() == b.params_flat[off:].data_ptr()
+                assert torch.equal(p.data, originals[name])  # repack kept values
+        assert seen == set(originals)
+
+    check()
+
+
 def test_grads_accumulate_into_flat():
     m = mnist_mlp(hidden=16)
     eng = SyncEngine(m, Comm(), bucket_cap_mb=0.01)
