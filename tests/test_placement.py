@@ -49,3 +49,32 @@ def test_replica_device_setter_shim():
     # no ps → everything on shard 0
     setter0 = replica_device_setter(ps_tasks=0)
     assert setter0("v", t) == 0
+
+
+def test_placement_fuzz_invariants():
+    """Property fuzz: both strategies assign every tensor exactly once;
+    greedy's max shard load never exceeds min load + the largest
+    tensor (the classic LPT-greedy bound for online assignment)."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(deadline=None, max_examples=40)
+    @given(
+        sizes=st.lists(st.integers(1, 10_000), min_size=1, max_size=40),
+        shards=st.integers(1, 8),
+        strategy=st.sampled_from(["round_robin", "greedy"]),
+    )
+    def check(sizes, shards, strategy):
+        named = [(f"v{i}", n) for i, n in enumerate(sizes)]
+        plan = plan_shards(named, shards, strategy=strategy)
+        assigned = [plan.shard_of(f"v{i}") for i in range(len(sizes))]
+        assert all(0 <= s < shards for s in assigned)
+        per_names = [plan.names_on(s) for s in range(shards)]
+        flat = [n for ns in per_names for n in ns]
+        assert sorted(flat) == sorted(n for n, _ in named)  # exactly once
+        if strategy == "greedy":
+            loads = [0] * shards
+            for (n, b), s in zip(named, assigned):
+                loads[s] += b
+            assert max(loads) <= min(loads) + max(sizes)
+
+    check()
