@@ -114,6 +114,16 @@ def test_sync_two_workers_parity(free_ports, tmp_path):
         assert abs(avg - single[s]) < 1e-4, (s, avg, single[s])
 
 
+def test_sync_cluster_pure_python_server(free_ports, tmp_path, monkeypatch):
+    """The whole cluster also runs on the pure-Python control-plane
+    server (ROCPS_PURE_PY_SERVER=1) — the fallback when the native
+    extension is unavailable."""
+    monkeypatch.setenv("ROCPS_PURE_PY_SERVER", "1")
+    losses, _ = _run_cluster(free_ports, tmp_path, issync=True,
+                             n_workers=2, steps=8, batch=8)
+    assert len(losses[0]) == 8 and len(losses[1]) == 8
+
+
 def test_sync_chief_writes_checkpoint(free_ports, tmp_path):
     ckdir = tmp_path / "ckpt"
     _run_cluster(free_ports, tmp_path, issync=True, steps=12, batch=8,
