@@ -397,9 +397,33 @@ at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
   }
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(A.device());
   auto D = at::empty({M, N}, A.options());
-  launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), (int)M,
-                   (int)N, (int)K, (int)epilogue, cur_stream(A));
+  launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), nullptr,
+                   (int)M, (int)N, (int)K, (int)epilogue, cur_stream(A));
   return D;
+}
+
+// epilogue-3 variant: returns (gelu(A@W^T + b), pre-activation) so the
+// training backward reuses pre instead of recomputing it (ROADMAP §Perf 5)
+std::tuple<at::Tensor, at::Tensor> gemm_bf16_gelu_pre(at::Tensor A,
+                                                      at::Tensor W,
+                                                      at::Tensor bias) {
+  TORCH_CHECK(A.is_cuda() && A.dim() == 2 && A.is_contiguous());
+  TORCH_CHECK(W.is_cuda() && W.dim() == 2 && W.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16
+              && W.scalar_type() == at::kBFloat16);
+  long M = A.size(0), K = A.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
+              "gemm_bf16 needs M%128==0, N%128==0, K%64==0");
+  auto b32 = bias.to(at::kFloat).contiguous();
+  TORCH_CHECK(b32.numel() == N);
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(A.device());
+  auto D = at::empty({M, N}, A.options());
+  auto P = at::empty({M, N}, A.options());
+  launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), b32.data_ptr<float>(),
+                   bf16_ptr(D), bf16_ptr(P), (int)M, (int)N, (int)K, 3,
+                   cur_stream(A));
+  return {D, P};
 }
 
 // ---------------------------------------------------------------------------
@@ -464,6 +488,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_fwd", &ln_fwd);
   m.def("ln_bwd", &ln_bwd);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("gemm_bf16_gelu_pre", &gemm_bf16_gelu_pre);
   m.def("maxpool332_fwd", &maxpool332_fwd);
   m.def("maxpool332_bwd", &maxpool332_bwd);
 }

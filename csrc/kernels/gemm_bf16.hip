@@ -42,11 +42,14 @@ __device__ __forceinline__ float gelu_tanh(float x) {
   return 0.5f * x * (1.f + t);
 }
 
-template <int EPI>  // 0 = none, 1 = +bias, 2 = +bias +gelu
+template <int EPI>  // 0 = none, 1 = +bias, 2 = +bias +gelu,
+                    // 3 = +bias +gelu, pre-activation ALSO stored to P
+                    //     (training: backward reuses pre instead of the
+                    //      fp32 recompute chain — ROADMAP §Perf 5)
 __global__ __launch_bounds__(256)
 void gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
                       const float* __restrict__ bias, bf16* __restrict__ D,
-                      int M, int N, int K) {
+                      bf16* __restrict__ P, int M, int N, int K) {
   __shared__ char smem[4 * TILE_BYTES];  // [buf][A|B]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -146,7 +149,8 @@ void gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
       for (int r = 0; r < 4; r++) {
         int row = m0 + wm + i * 16 + (lane >> 4) * 4 + r;
         float v = acc[i][j][r] + bv;
-        if (EPI == 2) v = gelu_tanh(v);
+        if (EPI == 3) P[(long)row * N + col] = from_f32<bf16>(v);
+        if (EPI >= 2) v = gelu_tanh(v);
         D[(long)row * N + col] = from_f32<bf16>(v);
       }
     }
@@ -156,19 +160,23 @@ void gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
 }  // namespace
 
 void launch_gemm_bf16(const rbf16* A, const rbf16* W, const float* bias,
-                      rbf16* D, int M, int N, int K, int epilogue,
-                      hipStream_t st) {
+                      rbf16* D, rbf16* PRE, int M, int N, int K,
+                      int epilogue, hipStream_t st) {
   dim3 grid(N / BN, M / BM);
   const bf16* a = reinterpret_cast<const bf16*>(A);
   const bf16* w = reinterpret_cast<const bf16*>(W);
   bf16* d = reinterpret_cast<bf16*>(D);
-  if (epilogue == 2)
+  bf16* p = reinterpret_cast<bf16*>(PRE);
+  if (epilogue == 3)
+    hipLaunchKernelGGL((gemm_bf16_kernel<3>), grid, dim3(256), 0, st, a, w,
+                       bias, d, p, M, N, K);
+  else if (epilogue == 2)
     hipLaunchKernelGGL((gemm_bf16_kernel<2>), grid, dim3(256), 0, st, a, w,
-                       bias, d, M, N, K);
+                       bias, d, nullptr, M, N, K);
   else if (epilogue == 1)
     hipLaunchKernelGGL((gemm_bf16_kernel<1>), grid, dim3(256), 0, st, a, w,
-                       bias, d, M, N, K);
+                       bias, d, nullptr, M, N, K);
   else
     hipLaunchKernelGGL((gemm_bf16_kernel<0>), grid, dim3(256), 0, st, a, w,
-                       bias, d, M, N, K);
+                       bias, d, nullptr, M, N, K);
 }

@@ -112,8 +112,8 @@ void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
 
 // gemm_bf16.hip
 void launch_gemm_bf16(const rbf16* A, const rbf16* W, const float* bias,
-                      rbf16* D, int M, int N, int K, int epilogue,
-                      hipStream_t st);
+                      rbf16* D, rbf16* PRE, int M, int N, int K,
+                      int epilogue, hipStream_t st);
 
 // maxpool.hip (3x3 stride-2 pad-1, NHWC)
 void launch_maxpool332_fwd_f32(const float* x, float* y, unsigned char* idx,

@@ -53,3 +53,33 @@ def test_bert_tiny_train_step():
         eng.step(loss, opt)
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0]  # overfits the fixed batch
+
+
+def test_linear_gelu_train_grads_match_autograd():
+    """linear_gelu_train (fused epilogue-3 path staged for round 2):
+    the saved-pre backward matches plain autograd on the CPU reference
+    path (same math the GPU kernel implements)."""
+    import torch
+
+    from rocps.ops.gemm import linear_gelu_train
+
+    g = torch.Generator().manual_seed(3)
+    x = torch.randn(16, 32, generator=g, requires_grad=True)
+    w = torch.randn(24, 32, generator=g, requires_grad=True)
+    b = torch.randn(24, generator=g, requires_grad=True)
+
+    y = linear_gelu_train(x, w, b)
+    y.backward(torch.ones_like(y))
+    gx, gw, gb = x.grad.clone(), w.grad.clone(), b.grad.clone()
+
+    x2 = x.detach().clone().requires_grad_()
+    w2 = w.detach().clone().requires_grad_()
+    b2 = b.detach().clone().requires_grad_()
+    pre = x2 @ w2.t() + b2
+    c = 0.7978845608028654
+    ref = 0.5 * pre * (1 + torch.tanh(c * (pre + 0.044715 * pre ** 3)))
+    assert torch.allclose(y, ref, atol=1e-5)
+    ref.backward(torch.ones_like(ref))
+    assert torch.allclose(gx, x2.grad, atol=1e-4)
+    assert torch.allclose(gw, w2.grad, atol=1e-4)
+    assert torch.allclose(gb, b2.grad, atol=1e-4)

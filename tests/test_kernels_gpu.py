@@ -2,6 +2,8 @@
 reference (the CPU path of the same op). Run on an MI355X with
 ``pytest -m gpu``."""
 
+import os
+
 import pytest
 import torch
 
@@ -440,3 +442,24 @@ def test_maxpool332_bf16_forward_exact():
     y = MaxPool332()(x)
     ref = torch.nn.functional.max_pool2d(x, 3, stride=2, padding=1)
     assert torch.equal(y, ref)
+
+
+@pytest.mark.skipif(os.environ.get("ROCPS_TEST_STAGED") != "1",
+                    reason="staged round-2 kernel (set ROCPS_TEST_STAGED=1)")
+def test_gemm_bf16_gelu_pre_staged():
+    """Epilogue-3 (gelu + pre-activation out) — staged for round 2's
+    BERT training path; run with ROCPS_TEST_STAGED=1."""
+    from rocps.ops import hip_ext
+
+    g = torch.Generator().manual_seed(0)
+    M, N, K = 256, 128, 64
+    a = (torch.randn(M, K, generator=g) / 4).bfloat16().to(DEV)
+    w = (torch.randn(N, K, generator=g) / 4).bfloat16().to(DEV)
+    b = torch.randn(N, generator=g).to(DEV)
+    y, pre = hip_ext().gemm_bf16_gelu_pre(a, w, b)
+    ref_pre = (a.float() @ w.float().t() + b.float())
+    c = 0.7978845608028654
+    ref_y = 0.5 * ref_pre * (1 + torch.tanh(
+        c * (ref_pre + 0.044715 * ref_pre ** 3)))
+    assert torch.allclose(pre.float(), ref_pre, atol=0.1, rtol=0.02)
+    assert torch.allclose(y.float(), ref_y, atol=0.1, rtol=0.02)
