@@ -18,6 +18,8 @@ from __future__ import annotations
 
 from typing import Any, Dict, Hashable
 
+import os
+
 import torch
 
 __all__ = ["Optimizer", "GradientDescentOptimizer", "MomentumOptimizer",
@@ -29,6 +31,36 @@ def _use_fused(p: torch.Tensor) -> bool:
         return False
     from rocps.ops import fused_optim
     return fused_optim.available()
+
+
+_CPU_APPLY = None
+_EMPTY = torch.empty(0)
+
+
+def _cpu_fused(p: torch.Tensor, g: torch.Tensor):
+    """Fused single-pass C++ CPU apply (csrc/cpu/cpu_apply.cc) for the
+    async PS shard thread — torch's op-at-a-time path makes ~8 memory
+    passes per momentum apply, the fused loop ~5 (ROADMAP §Perf 4).
+    SGD/momentum are bit-identical to the torch ops (FMA rounding
+    mirrored); Adam is within 1 ulp/step (torch's SLEEF sqrt is not
+    correctly rounded). ROCPS_CPU_FUSED=0 disables."""
+    global _CPU_APPLY
+    if p.device.type != "cpu" or os.environ.get("ROCPS_CPU_FUSED") == "0":
+        return None
+    if not (p.is_contiguous() and g.is_contiguous()):
+        return None
+    if p.dtype not in (torch.float32, torch.bfloat16):
+        return None
+    if g.dtype not in (torch.float32, torch.bfloat16):
+        return None
+    if _CPU_APPLY is None:
+        try:
+            from rocps import _cpuapply
+            _CPU_APPLY = _cpuapply
+        except ImportError:
+            _CPU_APPLY = False
+    return _CPU_APPLY or None
+
 
 
 class Optimizer:
@@ -95,6 +127,11 @@ class GradientDescentOptimizer(Optimizer):
             from rocps.ops import fused_optim
             fused_optim.sgd(params, grads, m if "master" in st else None, self.lr)
             return
+        capp = _cpu_fused(params, grads)
+        if capp is not None:
+            capp.sgd(params, grads, st.get("master", _EMPTY), self.lr,
+                     "master" in st)
+            return
         m.add_(grads.to(m.dtype), alpha=-self.lr)
         self._writeback(st, params)
 
@@ -125,6 +162,12 @@ class MomentumOptimizer(Optimizer):
             from rocps.ops import fused_optim
             fused_optim.momentum(params, grads, m if "master" in st else None,
                                  v, self.lr, self.momentum, self.use_nesterov)
+            return
+        capp = _cpu_fused(params, grads)
+        if capp is not None:
+            capp.momentum(params, grads, st.get("master", _EMPTY), v,
+                          self.lr, self.momentum, self.use_nesterov,
+                          "master" in st)
             return
         g32 = grads.to(torch.float32)
         v.mul_(self.momentum).add_(g32)
@@ -167,6 +210,12 @@ class AdamOptimizer(Optimizer):
             from rocps.ops import fused_optim
             fused_optim.adam(params, grads, mst if "master" in st else None,
                              m, v, alpha, self.beta1, self.beta2, self.epsilon)
+            return
+        capp = _cpu_fused(params, grads)
+        if capp is not None:
+            capp.adam(params, grads, st.get("master", _EMPTY), m, v,
+                      alpha, self.beta1, self.beta2, self.epsilon,
+                      "master" in st)
             return
         g32 = grads.to(torch.float32)
         m.mul_(self.beta1).add_(g32, alpha=1 - self.beta1)

@@ -41,6 +41,16 @@ ext_modules = [
         extra_compile_args=["-O2", "-std=c++17"] + _SAN_FLAGS,
         extra_link_args=_SAN_FLAGS,
     ),
+    CppExtension(
+        name="rocps._cpuapply",
+        sources=["csrc/cpu/cpu_apply.cc"],
+        # -ffp-contract=off: bit-parity with torch's unfused CPU ops;
+        # -mavx2 baseline (safe on the GPU pool's hosts), OMP across
+        # the shard
+        extra_compile_args=["-O3", "-std=c++17", "-mavx2", "-fopenmp",
+                            "-ffp-contract=off"] + _SAN_FLAGS,
+        extra_link_args=["-fopenmp"] + _SAN_FLAGS,
+    ),
 ]
 
 _hip_sources = [

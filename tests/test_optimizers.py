@@ -103,3 +103,60 @@ def test_state_per_slot_isolated():
     opt.apply_flat(a, torch.ones(4), slot="A")
     opt.apply_flat(b, torch.full((4,), 2.0), slot="B")
     assert not torch.equal(opt._state["A"]["mom"], opt._state["B"]["mom"])
+
+
+def test_cpu_fused_apply_matches_torch_path(monkeypatch):
+    """csrc/cpu/cpu_apply.cc (async PS apply, ROADMAP §Perf 4): the
+    fused single-pass loops match the torch-op path — bitwise for
+    SGD/momentum (FMA rounding mirrored), ~1 ulp for Adam (torch's
+    SLEEF sqrt is not correctly rounded)."""
+    pytest.importorskip("rocps._cpuapply")
+    import rocps.train.optimizer as O
+
+    n = 4097  # odd size: exercises any vector tail
+    for make, exact in [
+        (lambda: O.GradientDescentOptimizer(0.17), True),
+        (lambda: O.MomentumOptimizer(0.1, 0.9), True),
+        (lambda: O.MomentumOptimizer(0.1, 0.9, use_nesterov=True), True),
+        (lambda: O.AdamOptimizer(0.01), False),
+    ]:
+        g = torch.Generator().manual_seed(0)
+        grads = torch.randn(n, generator=g)
+        results = {}
+        for mode in ("fused", "torch"):
+            monkeypatch.setenv("ROCPS_CPU_FUSED",
+                               "1" if mode == "fused" else "0")
+            opt = make()
+            if hasattr(opt, "_t"):
+                opt._t = 3
+            p = torch.linspace(-2, 2, n)
+            for _ in range(5):
+                opt.apply_flat(p, grads, slot=0)
+            results[mode] = p
+        if exact:
+            assert torch.equal(results["fused"], results["torch"]), make
+        else:
+            assert torch.allclose(results["fused"], results["torch"],
+                                  atol=1e-5, rtol=1e-5)
+
+
+def test_cpu_fused_apply_bf16_master(monkeypatch):
+    """bf16 shard + fp32 master through the fused CPU path equals the
+    torch-op master-weight path."""
+    pytest.importorskip("rocps._cpuapply")
+    import rocps.train.optimizer as O
+
+    n = 1031
+    g = torch.Generator().manual_seed(1)
+    grads = torch.randn(n, generator=g).bfloat16()
+    results = {}
+    for mode in ("fused", "torch"):
+        monkeypatch.setenv("ROCPS_CPU_FUSED",
+                           "1" if mode == "fused" else "0")
+        opt = O.MomentumOptimizer(0.2, 0.9)
+        p = torch.linspace(-1, 1, n).bfloat16()
+        for _ in range(6):
+            opt.apply_flat(p, grads, slot=0)
+        results[mode] = (p, opt._state[0]["master"].clone())
+    assert torch.equal(results["fused"][0], results["torch"][0])
+    assert torch.equal(results["fused"][1], results["torch"][1])
