@@ -78,7 +78,7 @@ Arrays unpack(at::Tensor& p, const at::Tensor& g, at::Tensor& master,
 
 #define LOOP_BODY(UPDATE)                                              \
   if (a.pf) {                                                          \
-    _Pragma("omp parallel for schedule(static) num_threads(apply_threads())")                       \
+    _Pragma("omp parallel for schedule(static) num_threads(apply_threads())") \
     for (int64_t i = 0; i < a.n; i++) {                                \
       float gi = a.gf ? a.gf[i] : bf2f(a.gb[i]);                       \
       float m = a.pf[i];                                               \
@@ -86,7 +86,7 @@ Arrays unpack(at::Tensor& p, const at::Tensor& g, at::Tensor& master,
       a.pf[i] = m;                                                     \
     }                                                                  \
   } else {                                                             \
-    _Pragma("omp parallel for schedule(static) num_threads(apply_threads())")                       \
+    _Pragma("omp parallel for schedule(static) num_threads(apply_threads())") \
     for (int64_t i = 0; i < a.n; i++) {                                \
       float gi = a.gf ? a.gf[i] : bf2f(a.gb[i]);                       \
       float m = a.master[i];                                           \

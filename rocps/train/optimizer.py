@@ -62,7 +62,6 @@ def _cpu_fused(p: torch.Tensor, g: torch.Tensor):
     return _CPU_APPLY or None
 
 
-
 class Optimizer:
     """Functional flat-apply optimizer with lazily-created per-slot state."""
 
