@@ -527,6 +527,76 @@ def test_semisync_checkpoint_resume(free_ports, tmp_path):
     assert step2 >= step1 + 10  # continued, not restarted
 
 
+def test_semisync_worker_restart_rejoins():
+    """Elastic rejoin in backup-worker mode: a worker that detaches
+    mid-training is replaced (same task_index); rounds keep completing
+    with the remaining K-satisfiable set and the replacement's pushes
+    are handled after re-attach."""
+    import threading
+    import time as _time
+
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    W, K = 3, 2
+    workers, ps, cluster = create_local_cluster(num_workers=W, num_ps=1)
+    try:
+        shard = AsyncPSShard.serve_on(ps[0], cluster)
+        errs = []
+
+        def run_worker(w, steps, start_evt=None):
+            try:
+                if start_evt is not None:
+                    start_evt.wait(60)
+                torch.manual_seed(3)
+                m = mnist_mlp(hidden=16)
+                eng = AsyncWorkerEngine(
+                    m, cluster, ControlClient(cluster.master_address), w,
+                    torch.device("cpu"), replicas_to_aggregate=K,
+                    flush_timeout=0.5)
+                eng.initialize(GradientDescentOptimizer(0.2))
+                g = torch.Generator().manual_seed(50 + w)
+                for _ in range(steps):
+                    x, y = synthetic_mnist_batch(16, generator=g)
+                    eng.zero_grad()
+                    eng.step(m.loss(x, y), GradientDescentOptimizer(0.2))
+                eng.detach()
+            except Exception as e:  # noqa: BLE001
+                errs.append((w, e))
+
+        crashed = threading.Event()
+
+        def run_crasher():
+            run_worker(2, 3)       # 3 steps, then "dies" (detach)
+            crashed.set()
+
+        replacement_go = threading.Event()
+        ts = [threading.Thread(target=run_worker, args=(0, 14)),
+              threading.Thread(target=run_worker, args=(1, 14)),
+              threading.Thread(target=run_crasher),
+              threading.Thread(target=run_worker,
+                               args=(2, 6, replacement_go))]
+        for t in ts:
+            t.start()
+        assert crashed.wait(60)
+        _time.sleep(0.2)
+        replacement_go.set()       # restart worker 2 (same index)
+        for t in ts:
+            t.join(120)
+        assert not errs, errs
+        # every push (incl. the replacement's 6) was observed
+        assert int(shard.seg.header[5]) == 14 + 14 + 3 + 6
+        assert shard.applied >= (14 + 14) // K  # rounds kept completing
+    finally:
+        for s in workers + ps:
+            s.stop()
+
+
 def test_semisync_subprocess_cluster(free_ports, tmp_path):
     """--issync with --replicas_to_aggregate=2 of 3 workers end-to-end
     through the reference's CLI surface."""
