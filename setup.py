@@ -47,7 +47,7 @@ ext_modules = [
         # -ffp-contract=off: bit-parity with torch's unfused CPU ops;
         # -mavx2 baseline (safe on the GPU pool's hosts), OMP across
         # the shard
-        extra_compile_args=["-O3", "-std=c++17", "-mavx2", "-fopenmp",
+        extra_compile_args=["-O3", "-std=c++17", "-mavx2", "-mfma", "-fopenmp",
                             "-ffp-contract=off"] + _SAN_FLAGS,
         extra_link_args=["-fopenmp"] + _SAN_FLAGS,
     ),
