@@ -43,12 +43,17 @@ void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor master, double lr,
   long n = p.numel();
   auto st = cur_stream(p);
   if (p.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(g.scalar_type() == at::kBFloat16,
+                "sgd: bf16 params need bf16 grads");
     TORCH_CHECK(!has_master || master.scalar_type() == at::kFloat);
     launch_sgd_bf16(bf16_ptr(p), bf16_ptr(g),
                     has_master ? master.data_ptr<float>() : nullptr,
                     (float)lr, n, has_master, st);
   } else {
     TORCH_CHECK(p.scalar_type() == at::kFloat, "sgd: fp32/bf16 only");
+    TORCH_CHECK(g.scalar_type() == p.scalar_type(),
+                "sgd: grad dtype must match param dtype in fp32 mode");
+    TORCH_CHECK(!has_master, "sgd: master weights only with bf16 params");
     launch_sgd_f32(p.data_ptr<float>(), g.data_ptr<float>(),
                    p.data_ptr<float>(), (float)lr, n, false, st);
   }
@@ -66,11 +71,18 @@ void fused_momentum(at::Tensor p, at::Tensor g, at::Tensor master,
   long n = p.numel();
   auto st = cur_stream(p);
   if (p.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(g.scalar_type() == at::kBFloat16,
+                "momentum: bf16 params need bf16 grads");
     launch_momentum_bf16(bf16_ptr(p), bf16_ptr(g),
                          has_master ? master.data_ptr<float>() : nullptr,
                          mom.data_ptr<float>(), (float)lr, (float)mu,
                          nesterov, n, has_master, st);
   } else {
+    TORCH_CHECK(p.scalar_type() == at::kFloat, "momentum: fp32/bf16 only");
+    TORCH_CHECK(g.scalar_type() == p.scalar_type(),
+                "momentum: grad dtype must match param dtype in fp32 mode");
+    TORCH_CHECK(!has_master,
+                "momentum: master weights only with bf16 params");
     launch_momentum_f32(p.data_ptr<float>(), g.data_ptr<float>(),
                         p.data_ptr<float>(), mom.data_ptr<float>(), (float)lr,
                         (float)mu, nesterov, n, false, st);
@@ -89,11 +101,17 @@ void fused_adam(at::Tensor p, at::Tensor g, at::Tensor master, at::Tensor m,
   long n = p.numel();
   auto st = cur_stream(p);
   if (p.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(g.scalar_type() == at::kBFloat16,
+                "adam: bf16 params need bf16 grads");
     launch_adam_bf16(bf16_ptr(p), bf16_ptr(g),
                      has_master ? master.data_ptr<float>() : nullptr,
                      m.data_ptr<float>(), v.data_ptr<float>(), (float)alpha,
                      (float)b1, (float)b2, (float)eps, n, has_master, st);
   } else {
+    TORCH_CHECK(p.scalar_type() == at::kFloat, "adam: fp32/bf16 only");
+    TORCH_CHECK(g.scalar_type() == p.scalar_type(),
+                "adam: grad dtype must match param dtype in fp32 mode");
+    TORCH_CHECK(!has_master, "adam: master weights only with bf16 params");
     launch_adam_f32(p.data_ptr<float>(), g.data_ptr<float>(),
                     p.data_ptr<float>(), m.data_ptr<float>(),
                     v.data_ptr<float>(), (float)alpha, (float)b1, (float)b2,

@@ -17,10 +17,13 @@ MI355X-native recast (single node, 8 GPU workers + host-DRAM PS):
   copy. No per-tensor RPC, no serialization — the copy engine moves
   the bytes over PCIe/IF while compute streams stay busy.
 - The PS apply thread spins on seq!=ack, applies the (fused CPU)
-  optimizer per worker push, bumps version + global_step. Staleness
-  accounting: each push carries the worker's local step; the shard
-  drops pushes older than ``version - staleness_bound`` when a bound
-  is configured (the SyncReplicasOptimizer stale-drop rule, C4/C5).
+  optimizer per worker push, bumps version + global_step (one bump
+  per APPLY, the reference's AssignAdd semantics). Staleness
+  accounting: each worker records the shard version it observed at
+  pull time (pver); a push is dropped when the shard has advanced
+  more than ``staleness_bound`` applies since that pull (the
+  SyncReplicasOptimizer stale-drop rule recast per-apply, C4/C5).
+  Dropped pushes are still acked so the worker re-pulls and rejoins.
 
 Wire bootstrap (control plane): the chief publishes a manifest and
 init params; each PS task serves ``init_shard`` and publishes its shm
@@ -151,6 +154,7 @@ class AsyncPSShard:
         server.register_handler("opt_state_put", self._on_opt_state_put)
         self.applied = 0
         self.dropped_stale = 0
+        self.total_pushes = 0
         self.staleness_bound: Optional[int] = None
         self.replicas_to_aggregate: Optional[int] = None
         self.flush_timeout = 2.0
@@ -284,6 +288,7 @@ class AsyncPSShard:
         return {"ok": True, "applied": self.applied,
                 "dropped_stale": self.dropped_stale,
                 "partial_rounds": self.partial_rounds,
+                "total_pushes": self.total_pushes,
                 "version": int(self.seg.header[1]) if self.seg else 0}, b""
 
     # -- the apply loop --------------------------------------------------
@@ -307,9 +312,15 @@ class AsyncPSShard:
                     seq, ack = int(h[L.seq_idx(w)]), int(h[L.ack_idx(w)])
                     if seq != ack:
                         version = int(h[1])
-                        # seq encodes (worker local step); stale-drop rule
+                        # stale-drop rule: drop when the shard advanced
+                        # more than staleness_bound APPLIES since this
+                        # worker's last pull (pver = shard version the
+                        # worker observed at pull time). The dropped
+                        # push is still acked, so the worker re-pulls
+                        # (refreshing pver) and its next push is fresh.
                         if (self.staleness_bound is not None
-                                and version - seq > self.staleness_bound):
+                                and version - int(h[L.pver_idx(w)])
+                                > self.staleness_bound):
                             self.dropped_stale += 1
                         else:
                             if hasattr(self.optimizer, "begin_step"):
@@ -318,7 +329,10 @@ class AsyncPSShard:
                                 seg.params, seg.grads[w], slot="shard")
                             self.applied += 1
                             h[1] = version + 1
-                        h[5] += 1  # global_step mirror
+                            # global_step advances once per APPLY (the
+                            # reference's AssignAdd after ApplyXXX)
+                            h[5] += 1
+                        self.total_pushes += 1
                         h[L.ack_idx(w)] = seq  # ack
                         progressed = True
                 if not progressed:
@@ -353,6 +367,10 @@ class AsyncPSShard:
             self.applied += 1
             h[1] += 1
             h[6] = int(h[6]) + 1  # new round
+            # one aggregated apply = one global_step, matching the
+            # reference SyncReplicasOptimizer (step-keyed hooks and LR
+            # schedules advance per round, not per contribution)
+            h[5] += 1
             accum.zero_()
             n_contrib = 0
             for pw in pending:  # token-queue release
@@ -372,7 +390,7 @@ class AsyncPSShard:
                     if seq == ack:
                         continue
                     progressed = True
-                    h[5] += 1  # global_step counts every worker step
+                    self.total_pushes += 1
                     if int(h[L.pver_idx(w)]) != rnd:
                         self.dropped_stale += 1   # stale round
                         h[L.ack_idx(w)] = seq     # release the straggler
@@ -635,7 +653,12 @@ class AsyncWorkerEngine:
 
     def pull(self) -> None:
         w = self.task_index
-        rounds = [int(seg.header[6]) for seg in self.segs]
+        # pver records the shard VERSION (header[1], count of applies)
+        # observed at pull time. Async staleness compares the current
+        # version against it; in semisync the version and the round
+        # counter (header[6]) advance in lockstep (both bumped once per
+        # apply_round), so the freshness check pver == round holds too.
+        rounds = [int(seg.header[1]) for seg in self.segs]
         if self._side is not None:
             with torch.cuda.stream(self._side):
                 for s, seg in enumerate(self.segs):

@@ -35,6 +35,7 @@ class _FusedBNAct(torch.autograd.Function):
         has_res = residual is not None
         ctx.relu = relu
         ctx.has_res = has_res
+        ctx.training = training
         if require_hip_for(x):
             assert x.dim() == 4
             xc = x.contiguous(memory_format=torch.channels_last)
@@ -88,6 +89,24 @@ class _FusedBNAct(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, mean, invstd, y = ctx.saved_tensors[:5]
         fold4c = ctx.saved_tensors[5] if len(ctx.saved_tensors) > 5 else None
+        if not ctx.training:
+            # Eval-mode (frozen-BN) backward: mean/var are CONSTANTS
+            # (running stats), so dx has no batch-statistic correction
+            # terms — dx = dy * gamma * invstd, relu/residual-masked.
+            # Cold path (fine-tuning with frozen BN), computed with
+            # plain tensor ops on whatever device x lives on.
+            dy32 = dy.float()
+            if ctx.relu:
+                dy32 = dy32 * (y > 0).float()
+            dres = dy32.to(dy.dtype) if ctx.has_res else None
+            bc = (None, slice(None), None, None)
+            xhat = (x.float() - mean[bc]) * invstd[bc]
+            dbeta = dy32.sum(dim=(0, 2, 3))
+            dgamma = (dy32 * xhat).sum(dim=(0, 2, 3))
+            dx = dy32 * (weight.float() * invstd)[bc]
+            return (dx.to(x.dtype), dres, dgamma.to(weight.dtype),
+                    dbeta.to(weight.dtype),
+                    None, None, None, None, None, None)
         if ctx.hip:
             dyc = dy.contiguous(memory_format=torch.channels_last)
             outs = hip_ext().bn_relu_bwd(

@@ -490,8 +490,11 @@ def _semisync_once():
         handled = (shard.applied * K
                    - shard.partial_rounds * K  # partial rounds had < K
                    + sum(1 for _ in [None]) * 0)
-        total_pushes = int(shard.seg.header[5])
+        total_pushes = shard.total_pushes
         assert total_pushes == W * steps
+        # global_step advances once per aggregated apply (reference
+        # SyncReplicasOptimizer semantics), not per contribution
+        assert int(shard.seg.header[5]) == shard.applied
         assert shard.applied >= 1
         assert shard.dropped_stale >= 1  # the straggler lost rounds
         # accounting: fresh contributions + stale drops == pushes
@@ -590,7 +593,7 @@ def test_semisync_worker_restart_rejoins():
             t.join(120)
         assert not errs, errs
         # every push (incl. the replacement's 6) was observed
-        assert int(shard.seg.header[5]) == 14 + 14 + 3 + 6
+        assert shard.total_pushes == 14 + 14 + 3 + 6
         assert shard.applied >= (14 + 14) // K  # rounds kept completing
     finally:
         for s in workers + ps:
@@ -606,3 +609,78 @@ def test_semisync_subprocess_cluster(free_ports, tmp_path):
     for i in range(3):
         assert len(losses[i]) >= 1
         assert all(l == l and l < 50 for l in losses[i])
+
+
+def test_async_staleness_bound_multiworker_not_starved():
+    """the async stale-drop rule must compare the
+    shard version against the version the worker observed at PULL time
+    (pver), not against the worker's local step. With the old rule,
+    W>1 made version - local_step grow without bound, so once past
+    staleness_bound every push from every worker was silently dropped
+    forever. With the fix, lockstep workers stay fresh (gap ~W-1) and
+    a worker that slept through many applies has its stale push
+    dropped ONCE, re-pulls, and rejoins."""
+    import threading
+    import time as _time
+
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    W, steps, bound = 2, 20, 4
+    workers, ps, cluster = create_local_cluster(num_workers=W, num_ps=1)
+    try:
+        shard = AsyncPSShard.serve_on(ps[0], cluster)
+        models = []
+        for _ in range(W):
+            torch.manual_seed(7)
+            models.append(mnist_mlp(hidden=16))
+        engines = [AsyncWorkerEngine(
+            models[w], cluster, ControlClient(cluster.master_address), w,
+            torch.device("cpu"), staleness_bound=bound)
+            for w in range(W)]
+        errs = []
+
+        def run(w):
+            try:
+                eng = engines[w]
+                opt = GradientDescentOptimizer(0.2)
+                eng.initialize(opt)
+                g = torch.Generator().manual_seed(70 + w)
+                for i in range(steps):
+                    if w == 1 and i == steps // 2:
+                        # sleep through > bound applies by worker 0 so
+                        # the NEXT push is genuinely stale
+                        _time.sleep(1.0)
+                    x, y = synthetic_mnist_batch(16, generator=g)
+                    eng.zero_grad()
+                    loss = models[w].loss(x, y)
+                    eng.step(loss, opt)
+            except Exception as e:  # noqa: BLE001
+                errs.append((w, e))
+
+        ts = [threading.Thread(target=run, args=(w,)) for w in range(W)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(120)
+        assert not errs, errs
+        assert shard.total_pushes == W * steps
+        # conservation: every push either applied or dropped
+        assert shard.applied + shard.dropped_stale == W * steps
+        # the old bug dropped EVERYTHING after ~bound steps; the fixed
+        # rule keeps lockstep workers fresh — the overwhelming majority
+        # of pushes apply, and in particular pushes near the END of the
+        # run still applied (version kept advancing)
+        assert shard.applied >= W * steps - 6, (
+            shard.applied, shard.dropped_stale)
+        assert int(shard.seg.header[1]) == shard.applied
+        # global_step counts applies (reference AssignAdd semantics)
+        assert int(shard.seg.header[5]) == shard.applied
+    finally:
+        for s in workers + ps:
+            s.stop()

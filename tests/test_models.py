@@ -57,3 +57,40 @@ def test_bert_base_shape():
     lbl = torch.randint(0, 30522, (2, 16))
     loss = m.loss(ids, lbl)
     assert torch.isfinite(loss)
+
+
+def test_bn_eval_mode_backward_matches_frozen_stats():
+    """eval-mode (frozen-BN) backward must treat
+    running mean/var as CONSTANTS — dx = dy * gamma * invstd with no
+    batch-statistic correction terms. Compare against autograd through
+    F.batch_norm(training=False)."""
+    import torch.nn.functional as F
+
+    from rocps.ops.norm import fused_bn_act
+
+    torch.manual_seed(0)
+    N, C, H, W = 4, 8, 5, 5
+    for relu in (False, True):
+        x = torch.randn(N, C, H, W, dtype=torch.float32, requires_grad=True)
+        w = torch.randn(C, requires_grad=True)
+        b = torch.randn(C, requires_grad=True)
+        rm = torch.randn(C)
+        rv = torch.rand(C) + 0.5
+        y = fused_bn_act(x, w, b, rm.clone(), rv.clone(),
+                         training=False, relu=relu)
+        gy = torch.randn_like(y)
+        y.backward(gy)
+
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        ref = F.batch_norm(x2, rm.clone(), rv.clone(), w2, b2,
+                           training=False, eps=1e-5)
+        if relu:
+            ref = ref.relu()
+        ref.backward(gy)
+
+        assert torch.allclose(y, ref, atol=1e-5), (relu,)
+        assert torch.allclose(x.grad, x2.grad, atol=1e-5), (relu,)
+        assert torch.allclose(w.grad, w2.grad, atol=1e-4), (relu,)
+        assert torch.allclose(b.grad, b2.grad, atol=1e-4), (relu,)
