@@ -394,8 +394,28 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
 // MFMA bf16 GEMM (C10): D = A @ W^T (+bias) (+gelu)
 // ---------------------------------------------------------------------------
 
+// variant: -1 auto (256-tile 8-phase pipeline where the shape allows,
+// else the 128-tile kernel), 1 force 128, 2 force 256 (microbench A/B)
+static void dispatch_gemm(const at::Tensor& A, const at::Tensor& W,
+                          const float* bptr, at::Tensor& D, rbf16* pre,
+                          long M, long N, long K, long epilogue,
+                          long variant) {
+  bool use256 = gemm256_supported((int)M, (int)N, (int)K);
+  if (variant == 1) use256 = false;
+  if (variant == 2)
+    TORCH_CHECK(use256, "shape not supported by the 256-tile kernel");
+  if (use256)
+    launch_gemm_bf16_256(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), pre,
+                         (int)M, (int)N, (int)K, (int)epilogue,
+                         cur_stream(A));
+  else
+    launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), pre,
+                     (int)M, (int)N, (int)K, (int)epilogue, cur_stream(A));
+}
+
 at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
-                     std::optional<at::Tensor> bias, long epilogue) {
+                     std::optional<at::Tensor> bias, long epilogue,
+                     long variant) {
   TORCH_CHECK(A.is_cuda() && A.dim() == 2 && A.is_contiguous());
   TORCH_CHECK(W.is_cuda() && W.dim() == 2 && W.is_contiguous());
   TORCH_CHECK(A.scalar_type() == at::kBFloat16
@@ -415,8 +435,7 @@ at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
   }
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(A.device());
   auto D = at::empty({M, N}, A.options());
-  launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), nullptr,
-                   (int)M, (int)N, (int)K, (int)epilogue, cur_stream(A));
+  dispatch_gemm(A, W, bptr, D, nullptr, M, N, K, epilogue, variant);
   return D;
 }
 
@@ -438,9 +457,7 @@ std::tuple<at::Tensor, at::Tensor> gemm_bf16_gelu_pre(at::Tensor A,
   c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(A.device());
   auto D = at::empty({M, N}, A.options());
   auto P = at::empty({M, N}, A.options());
-  launch_gemm_bf16(bf16_ptr(A), bf16_ptr(W), b32.data_ptr<float>(),
-                   bf16_ptr(D), bf16_ptr(P), (int)M, (int)N, (int)K, 3,
-                   cur_stream(A));
+  dispatch_gemm(A, W, b32.data_ptr<float>(), D, bf16_ptr(P), M, N, K, 3, -1);
   return {D, P};
 }
 
@@ -505,7 +522,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_relu_bwd", &bn_relu_bwd);
   m.def("ln_fwd", &ln_fwd);
   m.def("ln_bwd", &ln_bwd);
-  m.def("gemm_bf16", &gemm_bf16);
+  m.def("gemm_bf16", &gemm_bf16, py::arg("A"), py::arg("W"),
+        py::arg("bias"), py::arg("epilogue"),
+        py::arg("variant") = -1);
   m.def("gemm_bf16_gelu_pre", &gemm_bf16_gelu_pre);
   m.def("maxpool332_fwd", &maxpool332_fwd);
   m.def("maxpool332_bwd", &maxpool332_bwd);

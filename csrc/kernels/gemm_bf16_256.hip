@@ -107,15 +107,20 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   };
 
   // ---- fragment reads (ds_read_b128) --------------------------------
+  // Per phase ONE base address VGPR: fragment steps (f*2048) are
+  // instruction immediates, and the kk=1 address is the kk=0 address
+  // with byte-bit 6 flipped (kk*64 lives in bit 6, the XOR swizzle in
+  // bits 4-6, the row field above bit 7 — no carries), so it costs one
+  // v_xor instead of a second live-range (a spilled address VGPR here
+  // forces a compiler vmcnt(0) that drains the whole glds pipeline).
   const int xorv = (lane & 7) << 4;
-  auto rd = [&](int buf, int h, int pr, int kk) -> bf16x8 {
-    int b = (kk * 64 + ((lane >> 4) << 4)) ^ xorv;
-    return *(const bf16x8*)(smem + buf * BUF + h * HALF + pr * 128 + b);
-  };
+  const int rdb = (((lane >> 4) << 4) ^ xorv);
   // a-frags: pr = wr*64 + f*16 + (lane&15)   (f = 0..3)
   // b-frags: pr = wc*32 + j*16 + (lane&15)   (j = 0..1 within the half)
   const int apr = wr * 64 + (lane & 15);
   const int bpr = wc * 32 + (lane & 15);
+  const int abase = apr * 128 + rdb;  // + buf*BUF + h*HALF + f*2048 (^64 kk)
+  const int bbase = bpr * 128 + rdb;
 
   bf16x8 a0[4][2], a1[4][2], b0[2][2], b1[2][2];
   f32x4 acc[8][4] = {};
@@ -130,15 +135,17 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   VMCNT(8);   // A0(0), B0(0) landed
   BAR();
 #pragma unroll
-  for (int f = 0; f < 4; f++)
+  for (int f = 0; f < 4; f++) {
+    a0[f][0] = *(const bf16x8*)(smem + H_A0 * HALF + abase + f * 2048);
+    a0[f][1] = *(const bf16x8*)(smem + ((H_A0 * HALF + abase) ^ 64)
+                                + f * 2048);
+  }
 #pragma unroll
-    for (int kk = 0; kk < 2; kk++)
-      a0[f][kk] = rd(0, H_A0, apr + f * 16, kk);
-#pragma unroll
-  for (int j = 0; j < 2; j++)
-#pragma unroll
-    for (int kk = 0; kk < 2; kk++)
-      b0[j][kk] = rd(0, H_B0, bpr + j * 16, kk);
+  for (int j = 0; j < 2; j++) {
+    b0[j][0] = *(const bf16x8*)(smem + H_B0 * HALF + bbase + j * 2048);
+    b0[j][1] = *(const bf16x8*)(smem + ((H_B0 * HALF + bbase) ^ 64)
+                                + j * 2048);
+  }
   VMCNT(6);   // B1(0) landed
   BAR();
 
@@ -152,14 +159,24 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   // Macro, not a lambda: pointer-passing the fragment arrays would
   // spill them to scratch; every index below is a compile-time
   // constant after unrolling.
-#define READ_A(dst, buf, h)                        \
-  _Pragma("unroll") for (int f = 0; f < 4; f++)    \
-      _Pragma("unroll") for (int kk = 0; kk < 2; kk++) \
-          dst[f][kk] = rd(buf, h, apr + f * 16, kk)
-#define READ_B(dst, buf, h)                        \
-  _Pragma("unroll") for (int f = 0; f < 2; f++)    \
-      _Pragma("unroll") for (int kk = 0; kk < 2; kk++) \
-          dst[f][kk] = rd(buf, h, bpr + f * 16, kk)
+#define READ_A(dst, buf, h)                                              \
+  {                                                                      \
+    const char* p0 = smem + (buf) * BUF + (h) * HALF + abase;            \
+    const char* p1 = smem + (((buf) * BUF + (h) * HALF + abase) ^ 64);   \
+    _Pragma("unroll") for (int f = 0; f < 4; f++) {                      \
+      dst[f][0] = *(const bf16x8*)(p0 + f * 2048);                       \
+      dst[f][1] = *(const bf16x8*)(p1 + f * 2048);                       \
+    }                                                                    \
+  }
+#define READ_B(dst, buf, h)                                              \
+  {                                                                      \
+    const char* p0 = smem + (buf) * BUF + (h) * HALF + bbase;            \
+    const char* p1 = smem + (((buf) * BUF + (h) * HALF + bbase) ^ 64);   \
+    _Pragma("unroll") for (int f = 0; f < 2; f++) {                      \
+      dst[f][0] = *(const bf16x8*)(p0 + f * 2048);                       \
+      dst[f][1] = *(const bf16x8*)(p1 + f * 2048);                       \
+    }                                                                    \
+  }
 #define MFMA16(am, bm, mbase, nbase)                                    \
   __builtin_amdgcn_s_setprio(1);                                        \
   _Pragma("unroll") for (int i = 0; i < 4; i++)                         \

@@ -59,6 +59,7 @@ _hip_sources = [
     "csrc/kernels/softmax_xent.hip",
     "csrc/kernels/bn_relu.hip",
     "csrc/kernels/gemm_bf16.hip",
+    "csrc/kernels/gemm_bf16_256.hip",
     "csrc/kernels/maxpool.hip",
     "csrc/kernels/bias_act.hip",
     "csrc/kernels/layernorm.hip",

@@ -463,3 +463,50 @@ def test_gemm_bf16_gelu_pre_staged():
         c * (ref_pre + 0.044715 * ref_pre ** 3)))
     assert torch.allclose(pre.float(), ref_pre, atol=0.1, rtol=0.02)
     assert torch.allclose(y.float(), ref_y, atol=0.1, rtol=0.02)
+
+
+@pytest.mark.parametrize("shape", [(256, 256, 128), (512, 256, 256),
+                                   (768, 512, 384), (4096, 3072, 768)])
+def test_gemm256_vs_128_bitwise(shape):
+    """The 256-tile 8-phase kernel vs the 128-tile kernel: both
+    accumulate the same 16x16x32 MFMA chain in increasing-k order, so
+    the outputs must be BITWISE equal — any difference is a staging /
+    swizzle / fragment-map bug, not rounding."""
+    from rocps.ops import hip_ext
+    M, N, K = shape
+    torch.manual_seed(7)
+    a = (torch.randn(M, K, device=DEV) / K ** 0.25).bfloat16()
+    w = (torch.randn(N, K, device=DEV) / K ** 0.25).bfloat16()
+    d256 = hip_ext().gemm_bf16(a, w, None, 0, 2)
+    d128 = hip_ext().gemm_bf16(a, w, None, 0, 1)
+    torch.cuda.synchronize()
+    assert torch.equal(d256, d128), (
+        (d256 != d128).sum().item(), "mismatched elements")
+
+
+def test_gemm256_numerics_vs_fp32():
+    """256-tile kernel vs an fp32 reference on asymmetric operands
+    (guide G9), including the bias+GELU epilogue."""
+    from rocps.ops import hip_ext
+    M, N, K = 1024, 512, 640
+    torch.manual_seed(8)
+    a = (torch.randn(M, K, device=DEV) / K ** 0.25).bfloat16()
+    w = (torch.randn(N, K, device=DEV) / K ** 0.25).bfloat16()
+    b = torch.randn(N, device=DEV)
+    got = hip_ext().gemm_bf16(a, w, b, 2, 2)
+    torch.cuda.synchronize()
+    pre = a.float() @ w.float().t() + b
+    ref = torch.nn.functional.gelu(pre, approximate="tanh")
+    assert torch.allclose(got.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gemm256_deterministic():
+    from rocps.ops import hip_ext
+    M, N, K = 512, 512, 256
+    torch.manual_seed(9)
+    a = (torch.randn(M, K, device=DEV) / 4).bfloat16()
+    w = (torch.randn(N, K, device=DEV) / 4).bfloat16()
+    r1 = hip_ext().gemm_bf16(a, w, None, 0, 2)
+    r2 = hip_ext().gemm_bf16(a, w, None, 0, 2)
+    torch.cuda.synchronize()
+    assert torch.equal(r1, r2)

@@ -107,19 +107,24 @@ def bench_ln(dev, reps):
 
 
 def bench_gemm(dev, reps):
-    from rocps.ops.gemm import matmul_nt
+    from rocps.ops import hip_ext
+    ext = hip_ext()
     for (M, N, K) in [(4096, 4096, 4096), (8192, 768, 768),
-                      (8192, 3072, 768), (8192, 768, 3072)]:
+                      (8192, 3072, 768), (8192, 768, 3072),
+                      (8192, 8192, 8192)]:
         a = (torch.randn(M, K, device=dev) / 8).bfloat16()
         w = (torch.randn(N, K, device=dev) / 8).bfloat16()
-        us = timeit(lambda: matmul_nt(a, w), reps)
-        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
-        print(f"rocps gemm_bf16 {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
-              flush=True)
+        flop = 2.0 * M * N * K
+        us = timeit(lambda: ext.gemm_bf16(a, w, None, 0, 1), reps)
+        print(f"rocps gemm128  {M}x{N}x{K}: {us:9.1f} us "
+              f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
+        if M % 256 == 0 and N % 256 == 0 and K % 128 == 0:
+            us = timeit(lambda: ext.gemm_bf16(a, w, None, 0, 2), reps)
+            print(f"rocps gemm256  {M}x{N}x{K}: {us:9.1f} us "
+                  f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
         us = timeit(lambda: a @ w.t(), reps)
-        tf = 2.0 * M * N * K / (us * 1e-6) / 1e12
-        print(f"rocblas matmul  {M}x{N}x{K}: {us:9.1f} us  {tf:7.1f} TF/s",
-              flush=True)
+        print(f"rocblas matmul {M}x{N}x{K}: {us:9.1f} us "
+              f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
 
 
 def main():
