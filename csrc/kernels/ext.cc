@@ -401,6 +401,14 @@ static void dispatch_gemm(const at::Tensor& A, const at::Tensor& W,
                           long M, long N, long K, long epilogue,
                           long variant) {
   bool use256 = gemm256_supported((int)M, (int)N, (int)K);
+  // Measured heuristic (profiles/gemm_micro_r2*.txt): the 256-tile
+  // pipeline runs at 1 workgroup/CU, so it needs a grid that fills the
+  // 256 CUs and enough K-depth to amortize its 6-half prologue; on
+  // small grids or shallow K the 128-tile kernel (2-3 blocks/CU) wins.
+  if (use256 && variant == -1) {
+    long blocks = (M / 256) * (N / 256);
+    use256 = blocks >= 192 && K >= 1024;
+  }
   if (variant == 1) use256 = false;
   if (variant == 2)
     TORCH_CHECK(use256, "shape not supported by the 256-tile kernel");

@@ -9,7 +9,7 @@
 //
 // Pipeline: the K loop runs in phases of {issue 4-or-8 ds_read_b128 ||
 // stage one 16 KB half-tile with 2 global_load_lds_dwordx4 || counted
-// s_waitcnt vmcnt(6) || raw s_barrier || lgkmcnt(0) || 16 MFMA wrapped
+// s_waitcnt vmcnt(6) || raw s_barrier || 16 MFMA wrapped
 // in s_setprio(1)}.  A half-tile staged at phase p is guaranteed by the
 // vmcnt at p+3 and read at p+4, so 3-4 halves (6-8 glds) stay in
 // flight ACROSS the barriers — the counted-vmcnt pipelining that the
@@ -126,7 +126,6 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   f32x4 acc[8][4] = {};
 
 #define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
-#define LGKM0 asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
 #define BAR() __builtin_amdgcn_s_barrier()
 
   // ---- prologue: 6 halves (tiles 0 and 1), first reads --------------
@@ -187,10 +186,13 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
                       am[i][kk], bm[j][kk], acc[(mbase) + i][(nbase) + j], \
                       0, 0, 0);                                         \
   __builtin_amdgcn_s_setprio(0)
+// No explicit lgkmcnt here: each phase's ds_reads feed the NEXT
+// phase's MFMAs, and hipcc already inserts counted lgkmcnt waits for
+// its own ds_read->MFMA data deps — an explicit lgkmcnt(0) would drain
+// the just-issued reads instead of hiding them under this phase's MFMA.
 #define PHASE_TAIL() \
   VMCNT(6);          \
-  BAR();             \
-  LGKM0
+  BAR()
 
   for (int t = 0; t < nkt; t += 2) {
     // tile t (buffer 0)
@@ -217,7 +219,6 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
 #undef MFMA16
 #undef PHASE_TAIL
 #undef VMCNT
-#undef LGKM0
 #undef BAR
 
   // ---- epilogue: C/D map col = lane&15, row = (lane>>4)*4 + r -------

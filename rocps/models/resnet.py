@@ -20,6 +20,7 @@ from typing import List, Optional
 import torch
 import torch.nn as nn
 
+from rocps.ops.conv1x1 import Conv1x1
 from rocps.ops.norm import BatchNormAct2d
 from rocps.ops.pool import MaxPool332
 
@@ -31,7 +32,10 @@ def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
 
 
 def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+    # Conv1x1 runs the MFMA-GEMM train path on shapes where it measured
+    # faster than MIOpen (rocps/ops/conv1x1.py dispatch table);
+    # state_dict-compatible with nn.Conv2d.
+    return Conv1x1(cin, cout, stride=stride)
 
 
 class Bottleneck(nn.Module):

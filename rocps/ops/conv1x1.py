@@ -76,3 +76,111 @@ def conv1x1_nhwc(x: torch.Tensor, weight: torch.Tensor,
         out = out.to(x.dtype)
     return (out.reshape(n, h, w, cout).permute(0, 3, 1, 2)
             .contiguous(memory_format=torch.channels_last))
+
+
+# ---------------------------------------------------------------------------
+# Training path: Conv1x1 module with autograd (round-2 dispatch)
+# ---------------------------------------------------------------------------
+#
+# forward : out_rows = rows(N*Ho*Wo, Cin) @ W(Cout, Cin)^T     [MFMA GEMM]
+# dgrad   : dx_rows  = dy_rows(M, Cout) @ (W^T)(Cin, Cout)^T   [MFMA GEMM]
+#           (the weight transpose is a (Cout x Cin) copy — trivially small
+#           next to the activation tensors)
+# wgrad   : dW = dy_rows^T @ rows                              [rocBLAS TN]
+#           (contraction over M with both operands M-major; a plain
+#           library GEMM per the MI355X design rules)
+#
+# Dispatch is PER SHAPE: only (cin, cout, stride) combinations measured
+# faster than MIOpen's igemm asm (tools/bench_conv1x1.py, fwd+bwd) run
+# the GEMM path; everything else stays on F.conv2d.  ROCPS_CONV1X1=0
+# disables, =1 forces all shapes (benching).
+
+import os
+
+import torch.nn as nn
+import torch.nn.functional as F
+
+# (cin, cout, stride) -> GEMM path wins fwd+bwd vs MIOpen on MI355X at
+# batch 256 (profiles/conv1x1_shapes_*.txt).  Updated from measurement.
+_GEMM_SHAPES = set()
+
+
+def set_gemm_shapes(shapes) -> None:
+    _GEMM_SHAPES.clear()
+    _GEMM_SHAPES.update(shapes)
+
+
+def _nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """a (M,K) @ b (N,K)^T with the MFMA kernel when the tile contract
+    holds, rocBLAS otherwise (GPU), fp32 matmul on CPU."""
+    if a.is_cuda and a.dtype == torch.bfloat16:
+        m, k = a.shape
+        n = b.shape[0]
+        if m % 128 == 0 and n % 128 == 0 and k % 64 == 0 and hip_available():
+            from rocps.ops import hip_ext
+            return hip_ext().gemm_bf16(a, b, None, 0, -1)
+        return a @ b.t()
+    return (a.float() @ b.float().t()).to(a.dtype)
+
+
+class _Conv1x1Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, stride):
+        n, cin, h, wd = x.shape
+        cout = w.shape[0]
+        ctx.stride = stride
+        ctx.in_shape = x.shape
+        xs = x[:, :, ::stride, ::stride] if stride > 1 else x
+        xs = xs.contiguous(memory_format=torch.channels_last)
+        ho, wo = xs.shape[2], xs.shape[3]
+        rows = xs.permute(0, 2, 3, 1).reshape(n * ho * wo, cin)
+        wmat = w.reshape(cout, cin)
+        ctx.save_for_backward(rows, wmat)
+        out = _nt(rows, wmat)
+        return (out.reshape(n, ho, wo, cout).permute(0, 3, 1, 2)
+                .contiguous(memory_format=torch.channels_last))
+
+    @staticmethod
+    def backward(ctx, dy):
+        rows, wmat = ctx.saved_tensors
+        s = ctx.stride
+        n, cin, h, wd = ctx.in_shape
+        cout = wmat.shape[0]
+        dyr = (dy.contiguous(memory_format=torch.channels_last)
+               .permute(0, 2, 3, 1).reshape(-1, cout))
+        # dgrad: dx_rows[m, cin] = sum_cout dy[m, cout] * W[cout, cin]
+        dxr = _nt(dyr, wmat.t().contiguous())
+        # wgrad: contraction over the huge M dim — library GEMM
+        if rows.is_cuda:
+            dw = dyr.t() @ rows
+        else:
+            dw = (dyr.float().t() @ rows.float()).to(wmat.dtype)
+        ho = (h + s - 1) // s
+        wo = (wd + s - 1) // s
+        dxs = (dxr.reshape(n, ho, wo, cin).permute(0, 3, 1, 2)
+               .contiguous(memory_format=torch.channels_last))
+        if s > 1:
+            dx = torch.zeros(ctx.in_shape, dtype=dy.dtype, device=dy.device
+                             ).contiguous(memory_format=torch.channels_last)
+            dx[:, :, ::s, ::s] = dxs
+        else:
+            dx = dxs
+        return dx, dw.reshape(cout, cin, 1, 1).to(wmat.dtype), None
+
+
+class Conv1x1(nn.Conv2d):
+    """Drop-in nn.Conv2d(k=1, bias=False) that runs the MFMA-GEMM
+    train path on shapes where it measured faster than MIOpen
+    (tools/bench_conv1x1.py); state_dict-compatible with nn.Conv2d."""
+
+    def __init__(self, cin: int, cout: int, stride: int = 1):
+        super().__init__(cin, cout, 1, stride=stride, bias=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        mode = os.environ.get("ROCPS_CONV1X1", "auto")
+        if (mode != "0" and x.is_cuda and x.dtype == torch.bfloat16
+                and (mode == "1"
+                     or (self.in_channels, self.out_channels,
+                         self.stride[0]) in _GEMM_SHAPES)):
+            return _Conv1x1Fn.apply(x, self.weight, self.stride[0])
+        return super().forward(x)

@@ -44,3 +44,43 @@ def test_conv1x1_padding_math():
     assert p.shape == (128, 64)
     assert p[100:].abs().sum() == 0
     assert _pad_to(p, 0, 128) is p  # already aligned: no copy
+
+
+@pytest.mark.parametrize("shape,stride", [
+    ((2, 64, 8, 8), 1),
+    ((2, 128, 8, 8), 2),
+])
+def test_conv1x1_module_autograd_matches_conv2d(shape, stride):
+    """The Conv1x1 train path (GEMM fwd + GEMM dgrad + TN wgrad) against
+    autograd through F.conv2d, forward AND gradients, incl. the stride-2
+    scatter in dgrad."""
+    import os
+
+    from rocps.ops.conv1x1 import _Conv1x1Fn
+
+    g = torch.Generator().manual_seed(1)
+    n, c, h, wd = shape
+    cout = 64
+    x = torch.randn(n, c, h, wd, generator=g).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    w = torch.randn(cout, c, 1, 1, generator=g).requires_grad_(True)
+    y = _Conv1x1Fn.apply(x, w, stride)
+    gy = torch.randn(y.shape, generator=g)
+    y.backward(gy)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    ref = F.conv2d(x2, w2, stride=stride)
+    ref.backward(gy)
+    assert torch.allclose(y, ref, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_conv1x1_module_state_dict_compat():
+    from rocps.ops.conv1x1 import Conv1x1
+    m = Conv1x1(64, 128, stride=2)
+    ref = torch.nn.Conv2d(64, 128, 1, stride=2, bias=False)
+    ref.load_state_dict(m.state_dict())   # same keys/shapes
+    x = torch.randn(2, 64, 8, 8)
+    assert torch.allclose(m(x), ref(x), atol=1e-5)

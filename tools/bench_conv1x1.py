@@ -41,23 +41,52 @@ def timeit(fn, reps=30, warmup=10):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--train", action="store_true",
+                    help="time fwd+bwd (the dispatch decision metric)")
     args = ap.parse_args()
     torch.backends.cudnn.benchmark = True
-    from rocps.ops.conv1x1 import conv1x1_nhwc
+    from rocps.ops.conv1x1 import _Conv1x1Fn, conv1x1_nhwc
 
-    print(f"{'shape':>28} {'miopen us':>10} {'gemm us':>10} {'win':>5}")
+    winners = []
+    mode = "fwd+bwd" if args.train else "fwd"
+    print(f"{'shape':>28} {'miopen us':>10} {'gemm us':>10} {'win':>5}"
+          f"   ({mode})")
     for cin, cout, h, s in SHAPES:
         x = torch.randn(args.batch, cin, h, h, device="cuda",
                         dtype=torch.bfloat16).contiguous(
             memory_format=torch.channels_last)
         w = torch.randn(cout, cin, 1, 1, device="cuda",
                         dtype=torch.bfloat16)
-        t_ref = timeit(lambda: F.conv2d(x, w, stride=s))
-        t_gemm = timeit(lambda: conv1x1_nhwc(x, w, stride=s))
+        if args.train:
+            xr = x.requires_grad_(True)
+            wr = w.clone().requires_grad_(True)
+
+            def ref_step():
+                y = F.conv2d(xr, wr, stride=s)
+                y.backward(torch.ones_like(y))
+                xr.grad = wr.grad = None
+
+            def gemm_step():
+                y = _Conv1x1Fn.apply(xr, wr, s)
+                y.backward(torch.ones_like(y))
+                xr.grad = wr.grad = None
+
+            t_ref = timeit(ref_step)
+            t_gemm = timeit(gemm_step)
+        else:
+            t_ref = timeit(lambda: F.conv2d(x, w, stride=s))
+            t_gemm = timeit(lambda: conv1x1_nhwc(x, w, stride=s))
         flops = 2 * args.batch * (h // s) ** 2 * cin * cout
+        if args.train:
+            flops *= 3
+        win = t_gemm < t_ref
+        if win:
+            winners.append((cin, cout, s))
         print(f"{(cin, cout, h, s)!s:>28} {t_ref:10.1f} {t_gemm:10.1f} "
-              f"{'GEMM' if t_gemm < t_ref else 'ref':>5}"
+              f"{'GEMM' if win else 'ref':>5}"
               f"   ({flops / t_gemm / 1e6:.0f} TF/s gemm)")
+    print("dispatch set (paste into rocps/ops/conv1x1.py _GEMM_SHAPES):")
+    print(sorted(set(winners)))
 
 
 if __name__ == "__main__":
