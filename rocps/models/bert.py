@@ -49,20 +49,24 @@ class BertLayer(nn.Module):
         self.ln2 = LayerNorm(hidden)
 
     def _mlp(self, x):
-        # Inference: fc1+bias+GELU run as ONE fused MFMA kernel (C10's
-        # BERT epilogue variant) — beats the rocBLAS GEMM + separate
-        # GELU pass on the (B*S,3072,768) shape. Training keeps the
-        # library GEMM + torch GELU: its backward reuses the saved
-        # pre-activation, whereas the fused op would recompute it
-        # (measured net loss in tokens/s).
-        if (x.is_cuda and x.dtype == torch.bfloat16
-                and not torch.is_grad_enabled()):
-            from rocps.ops.gemm import matmul_nt, supported
+        # fc1+bias+GELU run as ONE fused MFMA kernel (C10's BERT
+        # epilogue variant).  Inference uses the recompute-free
+        # epilogue; training uses the epilogue-3 variant that emits the
+        # pre-activation as a second output so backward reuses it
+        # instead of the fp32 recompute chain (ROADMAP §Perf 5; the
+        # round-1 recompute variant lost 9 ms/step).
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            from rocps.ops.gemm import (linear_gelu_train, matmul_nt,
+                                        supported)
             B, S, H = x.shape
             inter = self.fc1.out_features
             if supported(B * S, inter, H):
-                h = matmul_nt(x.reshape(B * S, H), self.fc1.weight,
-                              self.fc1.bias, gelu=True)
+                if torch.is_grad_enabled():
+                    h = linear_gelu_train(x.reshape(B * S, H),
+                                          self.fc1.weight, self.fc1.bias)
+                else:
+                    h = matmul_nt(x.reshape(B * S, H), self.fc1.weight,
+                                  self.fc1.bias, gelu=True)
                 return self.fc2(h.reshape(B, S, inter))
         return self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
 
