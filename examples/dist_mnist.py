@@ -46,6 +46,10 @@ flags.DEFINE_integer("seed", 42, "data/model seed")
 flags.DEFINE_string("loss_log", "", "write per-step losses (JSON) here")
 flags.DEFINE_integer("kill_after_step", 0,
                      "fault injection: crash this worker after N local steps")
+flags.DEFINE_boolean("elastic", False,
+                     "sync mode: reform the collective group from the "
+                     "surviving workers on peer death and restore the "
+                     "latest checkpoint (rocps.train.recovery)")
 FLAGS = flags.FLAGS
 
 
@@ -96,16 +100,37 @@ def main(_):
             save_checkpoint_steps=(FLAGS.save_checkpoint_steps or None),
             save_checkpoint_secs=(None if FLAGS.save_checkpoint_steps else 600),
     ) as sess:
+        recovery = None
+        if FLAGS.elastic and FLAGS.issync:
+            from rocps.train.recovery import ElasticRecovery
+            recovery = ElasticRecovery(
+                trainer, FLAGS.checkpoint_dir or None,
+                join_window=6.0, comm_timeout=30.0, dead_threshold=5.0)
         local_step = 0
         while not sess.should_stop():
             # deterministic synthetic batch for this global step; each
-            # worker takes its slice (the reference feeds next_batch())
+            # worker takes its slice (the reference feeds next_batch());
+            # under elastic recovery the slice follows the CURRENT rank
+            # (the surviving workers re-cover the batch space)
             g = torch.Generator().manual_seed(FLAGS.seed + trainer.global_step)
             x, y = synthetic_mnist_batch(global_batch, generator=g)
-            lo = FLAGS.task_index * FLAGS.batch_size
+            lo = trainer.comm.rank * FLAGS.batch_size
             xb = x[lo:lo + FLAGS.batch_size].to(trainer.device)
             yb = y[lo:lo + FLAGS.batch_size].to(trainer.device)
-            loss = sess.run(lambda: trainer.model.loss(xb, yb))
+            try:
+                if recovery is not None:
+                    recovery.check_peers()
+                loss = sess.run(lambda: trainer.model.loss(xb, yb))
+            except Exception as e:  # noqa: BLE001 — peer loss mid-step
+                if recovery is None:
+                    raise
+                print(f"[elastic] step failed ({type(e).__name__}: {e}); "
+                      f"reforming group", flush=True)
+                recovery.recover()
+                print(f"[elastic] regrouped: rank {trainer.comm.rank}/"
+                      f"{trainer.comm.world_size} at step "
+                      f"{trainer.global_step}", flush=True)
+                continue
             losses.append(float(loss))
             local_step += 1
             if FLAGS.kill_after_step and local_step >= FLAGS.kill_after_step:

@@ -101,8 +101,11 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 # (cin, cout, stride) -> GEMM path wins fwd+bwd vs MIOpen on MI355X at
-# batch 256 (profiles/conv1x1_shapes_*.txt).  Updated from measurement.
-_GEMM_SHAPES = set()
+# batch 256 (profiles/conv1x1_train_r2.txt): the 7x7-stage projections,
+# where the small spatial M keeps the rocBLAS TN wgrad cheap.  On the
+# large-M shapes MIOpen's dedicated wrw asm kernels win the backward;
+# a hand-written split-M wgrad kernel is the open item.
+_GEMM_SHAPES = {(512, 2048, 1), (2048, 512, 1)}
 
 
 def set_gemm_shapes(shapes) -> None:

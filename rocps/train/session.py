@@ -82,6 +82,11 @@ class Trainer:
                 comm = Comm(device=self.device)
         self.comm = comm
         self.mode = mode
+        # remembered so elastic recovery can rebuild the engine for a
+        # shrunken world (rocps/train/recovery.py)
+        self._engine_cfg = dict(sync_mode=sync_mode,
+                                bucket_cap_mb=bucket_cap_mb,
+                                grad_dtype=grad_dtype)
         if mode == "sync":
             self.engine = SyncEngine(self.model, comm, mode=sync_mode,
                                      bucket_cap_mb=bucket_cap_mb,
