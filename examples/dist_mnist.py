@@ -124,9 +124,20 @@ def main(_):
             except Exception as e:  # noqa: BLE001 — peer loss mid-step
                 if recovery is None:
                     raise
-                print(f"[elastic] step failed ({type(e).__name__}: {e}); "
-                      f"reforming group", flush=True)
-                recovery.recover()
+                print(f"[elastic] step failed ({type(e).__name__}: "
+                      f"{str(e)[:200]}); reforming group", flush=True)
+                err = None
+                for _ in range(3):
+                    try:
+                        recovery.recover()
+                        err = None
+                        break
+                    except Exception as e2:  # noqa: BLE001
+                        err = e2
+                        print(f"[elastic] recover attempt failed: "
+                              f"{str(e2)[:200]}", flush=True)
+                if err is not None:
+                    raise err
                 print(f"[elastic] regrouped: rank {trainer.comm.rank}/"
                       f"{trainer.comm.world_size} at step "
                       f"{trainer.global_step}", flush=True)

@@ -41,55 +41,74 @@ class WorkerLost(RuntimeError):
 class ElasticRecovery:
     def __init__(self, trainer, checkpoint_dir: Optional[str],
                  join_window: float = 8.0, comm_timeout: float = 60.0,
-                 dead_threshold: float = 6.0):
+                 dead_threshold: float = 6.0, join_cap: float = 45.0):
         self.trainer = trainer
         self.checkpoint_dir = checkpoint_dir
         self.join_window = join_window
         self.comm_timeout = comm_timeout
         self.dead_threshold = dead_threshold
-        self.generation = 0
+        self.join_cap = join_cap
+        # tasks in the CURRENT collective group; peers that died in an
+        # earlier generation stay excluded (no restart-rejoin in sync
+        # mode — matching the reference, a restarted worker comes back
+        # through a full job restart)
+        self.members = set(trainer.cluster.task_indices("worker"))
 
     # ------------------------------------------------------------------
     def check_peers(self) -> None:
-        """Cheap pre-step probe: raise WorkerLost if a worker went
-        silent (so ranks not yet inside the collective don't enter it)."""
+        """Cheap pre-step probe: raise WorkerLost if a CURRENT group
+        member went silent (so ranks not yet inside the collective
+        don't block on it)."""
         t = self.trainer
         if t.heartbeat is None:
             return
-        dead = [m for m in t.heartbeat.dead_members(self.dead_threshold)
-                if m.startswith("worker:")]
-        mine = f"worker:{t.task_index}"
-        if any(d != mine for d in dead):
-            raise WorkerLost(f"silent peers: {dead}")
+        dead = {int(m.split(":")[1])
+                for m in t.heartbeat.dead_members(self.dead_threshold)
+                if m.startswith("worker:")}
+        lost = (dead & self.members) - {t.task_index}
+        if lost:
+            raise WorkerLost(f"silent peers: {sorted(lost)}")
 
     # ------------------------------------------------------------------
-    def _roll_call(self) -> List[int]:
-        """Post presence under the new generation, then collect every
-        survivor that posts within the join window."""
+    def _dead(self) -> set:
+        t = self.trainer
+        if t.heartbeat is None:
+            return set()
+        return {int(m.split(":")[1])
+                for m in t.heartbeat.dead_members(self.dead_threshold)
+                if m.startswith("worker:")}
+
+    def _roll_call(self, epoch: int) -> List[int]:
+        """Post presence under the agreed epoch, then wait until every
+        current member has either posted or been declared dead (capped;
+        a peer that is alive per heartbeats is WAITED FOR — its broken
+        collective may take seconds to surface)."""
         t = self.trainer
         c = t.control
-        c.put(f"regroup/{self.generation}/task{t.task_index}", 1)
-        expected = list(t.cluster.task_indices("worker"))
-        deadline = time.time() + self.join_window
+        c.put(f"regroup/{epoch}/task{t.task_index}", 1)
+        expected = sorted(self.members)
+        deadline = time.time() + self.join_cap
+        quiesce = time.time() + self.join_window
         alive = {t.task_index}
-        while time.time() < deadline:
-            dead = set()
-            if t.heartbeat is not None:
-                dead = {int(m.split(":")[1])
-                        for m in t.heartbeat.dead_members(self.dead_threshold)
-                        if m.startswith("worker:")}
+        while True:
+            dead = self._dead()
             for i in expected:
                 if i in alive:
                     continue
                 try:
-                    c.get(f"regroup/{self.generation}/task{i}",
-                          blocking=False)
+                    c.get(f"regroup/{epoch}/task{i}", blocking=False)
                     alive.add(i)
                 except KeyError:
                     pass
+            now = time.time()
             if all(i in alive or i in dead for i in expected):
                 break
+            if now > deadline:
+                break
             time.sleep(0.2)
+            # settle at least join_window so stragglers make the room
+            if now < quiesce:
+                continue
         return sorted(alive)
 
     # ------------------------------------------------------------------
@@ -100,19 +119,24 @@ class ElasticRecovery:
         from rocps.train.saver import latest_checkpoint
 
         t = self.trainer
-        self.generation += 1
         try:
             if dist.is_initialized():
                 dist.destroy_process_group()
         except Exception:
             pass
 
-        alive = self._roll_call()
+        # epoch agreement through the control plane: every survivor of
+        # the SAME failure reads the same counter; the leader bumps it
+        # only after the assignment is published, so a straggler that
+        # enters recovery late still lands in the same room
+        epoch = t.control.read_counter("regroup_epoch") + 1
+        self.generation = epoch
+        alive = self._roll_call(epoch)
         if len(alive) < 1 or t.task_index not in alive:
             raise WorkerLost("roll call failed during regroup")
         new_rank = alive.index(t.task_index)
         world = len(alive)
-        key = f"regroup/{self.generation}/master"
+        key = f"regroup/{epoch}/master"
         if new_rank == 0:
             host = t.cluster.task_address(
                 "worker", t.task_index).rsplit(":", 1)[0]
@@ -120,8 +144,10 @@ class ElasticRecovery:
                 host = "127.0.0.1"
             addr = f"{host}:{_free_port()}"
             t.control.put(key, addr)
+            t.control.put_counter("regroup_epoch", epoch)
         else:
             addr = t.control.get(key, timeout=self.comm_timeout)
+        self.members = set(alive)
 
         backend = t.comm.backend
         device = t.comm.device
