@@ -447,6 +447,18 @@ at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
   return D;
 }
 
+at::Tensor gelu_bwd(at::Tensor pre, at::Tensor dy) {
+  TORCH_CHECK(pre.is_cuda() && pre.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dy.sizes() == pre.sizes()
+              && dy.scalar_type() == at::kBFloat16);
+  auto p = pre.contiguous(), d = dy.contiguous();
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(p.device());
+  auto g = at::empty_like(p);
+  launch_gelu_bwd(bf16_ptr(p), bf16_ptr(d), bf16_ptr(g), p.numel(),
+                  cur_stream(p));
+  return g;
+}
+
 // epilogue-3 variant: returns (gelu(A@W^T + b), pre-activation) so the
 // training backward reuses pre instead of recomputing it (ROADMAP §Perf 5)
 std::tuple<at::Tensor, at::Tensor> gemm_bf16_gelu_pre(at::Tensor A,
@@ -534,6 +546,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias"), py::arg("epilogue"),
         py::arg("variant") = -1);
   m.def("gemm_bf16_gelu_pre", &gemm_bf16_gelu_pre);
+  m.def("gelu_bwd", &gelu_bwd);
   m.def("maxpool332_fwd", &maxpool332_fwd);
   m.def("maxpool332_bwd", &maxpool332_bwd);
 }

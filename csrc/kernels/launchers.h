@@ -110,6 +110,10 @@ void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
                                  long M, float* abd, rbf16* dgamma,
                                  rbf16* dbeta, int C, hipStream_t st);
 
+// gelu.hip
+void launch_gelu_bwd(const rbf16* pre, const rbf16* dy, rbf16* g, long n,
+                     hipStream_t st);
+
 // gemm_bf16_256.hip (the 8-phase 256x256 pipeline)
 bool gemm256_supported(int M, int N, int K);
 void launch_gemm_bf16_256(const rbf16* A, const rbf16* W, const float* bias,

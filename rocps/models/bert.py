@@ -55,7 +55,9 @@ class BertLayer(nn.Module):
         # pre-activation as a second output so backward reuses it
         # instead of the fp32 recompute chain (ROADMAP §Perf 5; the
         # round-1 recompute variant lost 9 ms/step).
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        import os
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and os.environ.get("ROCPS_BERT_FUSED", "1") != "0"):
             from rocps.ops.gemm import (linear_gelu_train, matmul_nt,
                                         supported)
             B, S, H = x.shape

@@ -96,7 +96,14 @@ class _LinearGeluTrainFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w, b, pre = ctx.saved_tensors
-        g = (dy.float() * _dgelu_tanh(pre.float())).to(dy.dtype)
+        if dy.is_cuda and dy.dtype == torch.bfloat16:
+            # one-pass fused dy * dgelu(pre) (csrc/kernels/gelu.hip) —
+            # the composed torch chain materializes several fp32
+            # (B*S, inter) intermediates per layer
+            from rocps.ops import hip_ext
+            g = hip_ext().gelu_bwd(pre, dy.contiguous())
+        else:
+            g = (dy.float() * _dgelu_tanh(pre.float())).to(dy.dtype)
         dx = g @ w
         dw = g.t() @ x
         db = g.float().sum(0)

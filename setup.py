@@ -58,6 +58,7 @@ _hip_sources = [
     "csrc/kernels/fused_optimizers.hip",
     "csrc/kernels/softmax_xent.hip",
     "csrc/kernels/bn_relu.hip",
+    "csrc/kernels/gelu.hip",
     "csrc/kernels/gemm_bf16.hip",
     "csrc/kernels/gemm_bf16_256.hip",
     "csrc/kernels/maxpool.hip",
