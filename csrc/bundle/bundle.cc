@@ -379,11 +379,26 @@ static BlockHandle emit_block(std::ofstream& f, uint64_t* file_off,
 // ---------------------------------------------------------------------------
 class BundleWriter {
  public:
-  explicit BundleWriter(const std::string& prefix) : prefix_(prefix) {
-    data_path_ = prefix_ + ".data-00000-of-00001";
-    tmp_data_path_ = data_path_ + ".tempstate";
-    data_.open(tmp_data_path_, std::ios::binary | std::ios::trunc);
-    if (!data_) throw std::runtime_error("cannot open " + tmp_data_path_);
+  // num_shards > 1 writes .data-SSSSS-of-NNNNN files with tensors
+  // assigned greedily to the least-loaded shard (the reference's
+  // sharded SaveV2 produces the same multi-file layout; readers find
+  // each tensor through its entry's shard_id).
+  explicit BundleWriter(const std::string& prefix, int num_shards = 1)
+      : prefix_(prefix), num_shards_(num_shards) {
+    if (num_shards < 1 || num_shards > 99999)
+      throw std::runtime_error("num_shards out of range");
+    data_.resize(num_shards);
+    offsets_.assign(num_shards, 0);
+    for (int sdx = 0; sdx < num_shards; sdx++) {
+      char nm[64];
+      snprintf(nm, sizeof(nm), ".data-%05d-of-%05d", sdx, num_shards);
+      data_paths_.push_back(prefix_ + nm);
+      tmp_data_paths_.push_back(prefix_ + nm + ".tempstate");
+      data_[sdx].open(tmp_data_paths_[sdx],
+                      std::ios::binary | std::ios::trunc);
+      if (!data_[sdx])
+        throw std::runtime_error("cannot open " + tmp_data_paths_[sdx]);
+    }
   }
 
   void add(const std::string& name, int dtype,
@@ -393,26 +408,33 @@ class BundleWriter {
     py::buffer_info info = buf.request();
     const char* ptr = static_cast<const char*>(info.ptr);
     size_t nbytes = static_cast<size_t>(info.size) * info.itemsize;
+    // greedy least-loaded shard
+    int sdx = 0;
+    for (int i = 1; i < num_shards_; i++)
+      if (offsets_[i] < offsets_[sdx]) sdx = i;
     BundleEntry e;
     e.dtype = dtype;
     e.shape = shape;
-    e.shard_id = 0;
-    e.offset = static_cast<int64_t>(offset_);
+    e.shard_id = sdx;
+    e.offset = static_cast<int64_t>(offsets_[sdx]);
     e.size = static_cast<int64_t>(nbytes);
     e.crc32c_masked = crc32c::mask(crc32c::value(ptr, nbytes));
     {
       py::gil_scoped_release rel;
-      data_.write(ptr, nbytes);
+      data_[sdx].write(ptr, nbytes);
     }
-    offset_ += nbytes;
+    offsets_[sdx] += nbytes;
     entries_[name] = e;
   }
 
   void finish() {
     if (finished_) return;
-    data_.close();
-    if (std::rename(tmp_data_path_.c_str(), data_path_.c_str()) != 0)
-      throw std::runtime_error("rename failed for " + data_path_);
+    for (int sdx = 0; sdx < num_shards_; sdx++) {
+      data_[sdx].close();
+      if (std::rename(tmp_data_paths_[sdx].c_str(),
+                      data_paths_[sdx].c_str()) != 0)
+        throw std::runtime_error("rename failed for " + data_paths_[sdx]);
+    }
 
     std::string index_path = prefix_ + ".index";
     std::string tmp_index = index_path + ".tempstate";
@@ -422,7 +444,7 @@ class BundleWriter {
 
     // Entries sorted; "" (header) first by std::map ordering.
     std::map<std::string, std::string> kv;
-    kv[""] = encode_header(1);
+    kv[""] = encode_header(num_shards_);
     for (auto& it : entries_) kv[it.first] = encode_entry(it.second);
 
     BlockBuilder data_block;
@@ -469,9 +491,11 @@ class BundleWriter {
   }
 
  private:
-  std::string prefix_, data_path_, tmp_data_path_;
-  std::ofstream data_;
-  uint64_t offset_ = 0;
+  std::string prefix_;
+  int num_shards_ = 1;
+  std::vector<std::string> data_paths_, tmp_data_paths_;
+  std::vector<std::ofstream> data_;
+  std::vector<uint64_t> offsets_;
   std::map<std::string, BundleEntry> entries_;
   bool finished_ = false;
 };
@@ -503,7 +527,22 @@ class BundleReader {
       if (!q) throw std::runtime_error("bad block handle");
       for (auto& kv : read_block(contents, h)) {
         if (kv.first.empty()) {
-          num_shards_ = 1;  // header; shard count re-derivable if needed
+          // BundleHeaderProto field 1 = num_shards (varint)
+          const char* hp = kv.second.data();
+          const char* he = hp + kv.second.size();
+          num_shards_ = 1;
+          while (hp < he) {
+            uint64_t tag, val;
+            if (!(hp = get_varint64(hp, he, &tag))) break;
+            if ((tag >> 3) == 1 && (tag & 7) == 0) {
+              if ((hp = get_varint64(hp, he, &val))) {
+                num_shards_ = static_cast<int>(val);
+              }
+              break;
+            }
+            break;  // other fields: stop (only num_shards matters)
+          }
+          if (num_shards_ < 1) num_shards_ = 1;
           continue;
         }
         BundleEntry e;
@@ -524,6 +563,8 @@ class BundleReader {
       d["dtype"] = it.second.dtype;
       d["shape"] = it.second.shape;
       d["size"] = it.second.size;
+      d["shard_id"] = it.second.shard_id;
+      d["offset"] = it.second.offset;
       out.append(d);
     }
     return out;
@@ -608,7 +649,8 @@ class BundleReader {
 PYBIND11_MODULE(_bundle, m) {
   m.doc() = "TF-V2 tensor-bundle checkpoint codec (rocps native)";
   py::class_<BundleWriter>(m, "BundleWriter")
-      .def(py::init<const std::string&>())
+      .def(py::init<const std::string&, int>(), py::arg("prefix"),
+           py::arg("num_shards") = 1)
       .def("add", &BundleWriter::add, py::arg("name"), py::arg("dtype"),
            py::arg("shape"), py::arg("data"))
       .def("finish", &BundleWriter::finish);

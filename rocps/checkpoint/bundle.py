@@ -43,8 +43,9 @@ def _to_bytes_view(t: torch.Tensor) -> np.ndarray:
     return t.numpy()
 
 
-def write_bundle(prefix: str, tensors: Dict[str, torch.Tensor]) -> None:
-    w = BundleWriter(prefix)
+def write_bundle(prefix: str, tensors: Dict[str, torch.Tensor],
+                 num_shards: int = 1) -> None:
+    w = BundleWriter(prefix, num_shards)
     for name in sorted(tensors):
         t = tensors[name]
         dt = TF_DTYPE.get(t.dtype)
