@@ -149,12 +149,16 @@ class AsyncPSShard:
         self.optimizer = None
         self._apply_thread: Optional[threading.Thread] = None
         server.register_handler("init_shard", self._on_init_shard)
+        server.register_handler("push_pull", self._on_push_pull)
         server.register_handler("shard_stats", self._on_stats)
         server.register_handler("opt_state_get", self._on_opt_state_get)
         server.register_handler("opt_state_put", self._on_opt_state_put)
         self.applied = 0
         self.dropped_stale = 0
         self.total_pushes = 0
+        # serializes optimizer applies between the shm apply loop and
+        # the TCP push_pull handler threads (remote workers)
+        self._apply_mu = threading.Lock()
         self.staleness_bound: Optional[int] = None
         self.replicas_to_aggregate: Optional[int] = None
         self.flush_timeout = 2.0
@@ -291,6 +295,48 @@ class AsyncPSShard:
                 "total_pushes": self.total_pushes,
                 "version": int(self.seg.header[1]) if self.seg else 0}, b""
 
+    def _on_push_pull(self, head, payload):
+        """Cross-host worker data path (the reference's PS
+        worked over TCP between hosts).  One round trip = gradient push
+        + immediate apply + fresh-parameter pull, i.e. the reference's
+        RunGraph exchange.  Remote workers use it when the shard's
+        /dev/shm segment is not reachable (different host); local
+        workers keep the shm fast path.  Async mode only — semisync
+        round membership needs the shm seq/ack slots."""
+        if self.seg is None:
+            return {"ok": False, "error": "shard uninitialized"}, b""
+        if self.replicas_to_aggregate:
+            return {"ok": False,
+                    "error": "tcp transport is async-mode only"}, b""
+        seg = self.seg
+        h = seg.header
+        with self._apply_mu:
+            if payload and not head.get("pull_only"):
+                version = int(h[1])
+                if (self.staleness_bound is not None
+                        and version - int(head.get("pver", 0))
+                        > self.staleness_bound):
+                    self.dropped_stale += 1
+                else:
+                    src = np.frombuffer(payload,
+                                        dtype=_np_dtype(seg.dtype),
+                                        count=seg.layout.numel)
+                    g = torch.from_numpy(src.copy())
+                    if seg.dtype == torch.bfloat16:
+                        g = g.view(torch.bfloat16)
+                    if hasattr(self.optimizer, "begin_step"):
+                        self.optimizer.begin_step()
+                    self.optimizer.apply_flat(seg.params, g, slot="shard")
+                    self.applied += 1
+                    h[1] = version + 1
+                    h[5] += 1
+                self.total_pushes += 1
+            view = (seg.params.view(torch.uint16)
+                    if seg.dtype == torch.bfloat16 else seg.params)
+            out = view.numpy().tobytes()
+            return {"ok": True, "version": int(h[1]),
+                    "global_step": int(h[5])}, out
+
     # -- the apply loop --------------------------------------------------
     def _apply_loop(self) -> None:
         if self.replicas_to_aggregate:
@@ -311,28 +357,29 @@ class AsyncPSShard:
                 for w in range(W):
                     seq, ack = int(h[L.seq_idx(w)]), int(h[L.ack_idx(w)])
                     if seq != ack:
-                        version = int(h[1])
                         # stale-drop rule: drop when the shard advanced
                         # more than staleness_bound APPLIES since this
                         # worker's last pull (pver = shard version the
                         # worker observed at pull time). The dropped
                         # push is still acked, so the worker re-pulls
                         # (refreshing pver) and its next push is fresh.
-                        if (self.staleness_bound is not None
-                                and version - int(h[L.pver_idx(w)])
-                                > self.staleness_bound):
-                            self.dropped_stale += 1
-                        else:
-                            if hasattr(self.optimizer, "begin_step"):
-                                self.optimizer.begin_step()
-                            self.optimizer.apply_flat(
-                                seg.params, seg.grads[w], slot="shard")
-                            self.applied += 1
-                            h[1] = version + 1
-                            # global_step advances once per APPLY (the
-                            # reference's AssignAdd after ApplyXXX)
-                            h[5] += 1
-                        self.total_pushes += 1
+                        with self._apply_mu:
+                            version = int(h[1])
+                            if (self.staleness_bound is not None
+                                    and version - int(h[L.pver_idx(w)])
+                                    > self.staleness_bound):
+                                self.dropped_stale += 1
+                            else:
+                                if hasattr(self.optimizer, "begin_step"):
+                                    self.optimizer.begin_step()
+                                self.optimizer.apply_flat(
+                                    seg.params, seg.grads[w], slot="shard")
+                                self.applied += 1
+                                h[1] = version + 1
+                                # global_step advances once per APPLY
+                                # (the reference's AssignAdd)
+                                h[5] += 1
+                            self.total_pushes += 1
                         h[L.ack_idx(w)] = seq  # ack
                         progressed = True
                 if not progressed:
@@ -497,6 +544,10 @@ class AsyncWorkerEngine:
             self.flat_params.append(fp)
             self.flat_grads.append(fg)
         self.segs: List[Optional[_ShmSeg]] = [None] * self.num_shards
+        # cross-host shards: ControlClient data path instead of shm
+        self._tcp: Dict[int, ControlClient] = {}
+        self._tcp_meta: Dict[int, tuple] = {}
+        self._tcp_pver: Dict[int, int] = {}
         # side stream: push/pull copies run off the compute stream (C7's
         # third stream) so pinned DMA overlaps across shards
         self._side = (torch.cuda.Stream(device=device)
@@ -575,6 +626,7 @@ class AsyncWorkerEngine:
         return list(r["keys"]), out, int(r.get("t", 0))
 
     def _attach(self) -> None:
+        force_tcp = os.environ.get("ROCPS_ASYNC_TCP") == "1"
         for s in range(self.num_shards):
             info = self.control.get(f"async_shm_{s}", blocking=True,
                                     timeout=120.0)
@@ -582,9 +634,26 @@ class AsyncWorkerEngine:
             itemsize = torch.empty(0, dtype=dtype).element_size()
             layout = _ShmLayout(int(info["numel"]),
                                 self.cluster.num_tasks("worker"), itemsize)
-            seg = _ShmSeg(info["path"], layout, dtype, create=False)
-            if self.device.type == "cuda":
-                self._pin_segment(seg, layout)
+            seg = None
+            if not force_tcp:
+                try:
+                    seg = _ShmSeg(info["path"], layout, dtype, create=False)
+                except OSError:
+                    seg = None  # shard on another host: TCP data path
+            if seg is None:
+                # cross-host transport (the reference's PS
+                # spoke TCP between hosts): push+apply+pull is one
+                # push_pull round trip to the shard's control server
+                ps_addr = self.cluster.task_address(
+                    "ps", self.cluster.task_indices("ps")[s])
+                self._tcp[s] = ControlClient(ps_addr)
+                self._tcp_meta[s] = (int(info["numel"]), dtype)
+                self._tcp_pver[s] = 0
+                if self.device.type == "cuda":
+                    pass  # params land via flat_params copy below
+            else:
+                if self.device.type == "cuda":
+                    self._pin_segment(seg, layout)
             self.segs[s] = seg
         self._attached = True
         self.pull()
@@ -619,9 +688,36 @@ class AsyncWorkerEngine:
         loss.backward()
         self.push_and_pull()
 
+    def _tcp_round_trip(self, s: int, push: bool) -> None:
+        numel, dtype = self._tcp_meta[s]
+        payload = b""
+        if push:
+            flat = self.flat_grads[s].detach().to("cpu")
+            view = (flat.view(torch.uint16) if dtype == torch.bfloat16
+                    else flat)
+            payload = view.numpy().tobytes()
+        r, params = self._tcp[s].call_raw(
+            "push_pull", payload=payload, worker=self.task_index,
+            pver=self._tcp_pver[s], pull_only=not push)
+        if not r.get("ok"):
+            raise RuntimeError(f"push_pull failed on shard {s}: {r}")
+        arr = np.frombuffer(params, dtype=_np_dtype(dtype), count=numel)
+        t = torch.from_numpy(arr.copy())
+        if dtype == torch.bfloat16:
+            t = t.view(torch.bfloat16)
+        self.flat_params[s].copy_(t.to(self.flat_params[s].device))
+        self._tcp_pver[s] = int(r["version"])
+        self.last_global_step = int(r["global_step"])
+
     def push_and_pull(self) -> None:
         self.local_step += 1
         w = self.task_index
+        if self._tcp:
+            for s in list(self._tcp):
+                self._tcp_round_trip(s, push=True)
+            if len(self._tcp) == self.num_shards:
+                return
+            # mixed shm/tcp: fall through for the shm shards only
         # push grads to every shard, bump seq (D2H straight into the
         # shard slot — pinned DMA on the side stream when _pin_segment
         # succeeded)
@@ -629,27 +725,37 @@ class AsyncWorkerEngine:
             self._side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self._side):
                 for s, seg in enumerate(self.segs):
+                    if seg is None:
+                        continue
                     seg.grads[w].copy_(self.flat_grads[s],
                                        non_blocking=getattr(seg, "pinned",
                                                             False))
             self._side.synchronize()  # grads visible before the seq bump
             for s, seg in enumerate(self.segs):
-                seg.header[_ShmLayout.seq_idx(w)] = self.local_step
+                if seg is not None:
+                    seg.header[_ShmLayout.seq_idx(w)] = self.local_step
         else:
             for s, seg in enumerate(self.segs):
+                if seg is None:
+                    continue
                 seg.grads[w].copy_(self.flat_grads[s])
                 seg.header[_ShmLayout.seq_idx(w)] = self.local_step
         # wait for own applies (the reference's RunGraph round trip; in
         # semi-sync mode this is the token-queue wait for the round)
         deadline = time.time() + 120.0
         for seg in self.segs:
+            if seg is None:
+                continue
             while int(seg.header[_ShmLayout.ack_idx(w)]) != self.local_step:
                 if time.time() > deadline:
                     raise TimeoutError("PS apply timeout (shard dead?)")
                 time.sleep(0.0002)
         # pull fresh (possibly concurrently-updated) params
         self.pull()
-        self.last_global_step = int(self.segs[0].header[5])
+        for seg in self.segs:
+            if seg is not None:
+                self.last_global_step = int(seg.header[5])
+                break
 
     def pull(self) -> None:
         w = self.task_index
@@ -658,21 +764,28 @@ class AsyncWorkerEngine:
         # version against it; in semisync the version and the round
         # counter (header[6]) advance in lockstep (both bumped once per
         # apply_round), so the freshness check pver == round holds too.
-        rounds = [int(seg.header[1]) for seg in self.segs]
+        rounds = [int(seg.header[1]) if seg is not None else 0
+                  for seg in self.segs]
+        for s in self._tcp:
+            self._tcp_round_trip(s, push=False)
         if self._side is not None:
             with torch.cuda.stream(self._side):
                 for s, seg in enumerate(self.segs):
+                    if seg is None:
+                        continue
                     self.flat_params[s].copy_(seg.params,
                                               non_blocking=getattr(
                                                   seg, "pinned", False))
             torch.cuda.current_stream().wait_stream(self._side)
         else:
             for s, seg in enumerate(self.segs):
-                self.flat_params[s].copy_(seg.params)
+                if seg is not None:
+                    self.flat_params[s].copy_(seg.params)
         # claim the round read BEFORE the copy: if an apply landed
         # mid-pull our grads will be (conservatively) counted stale
         for s, seg in enumerate(self.segs):
-            seg.header[_ShmLayout.pver_idx(w)] = rounds[s]
+            if seg is not None:
+                seg.header[_ShmLayout.pver_idx(w)] = rounds[s]
 
     def detach(self) -> None:
         for seg in self.segs:

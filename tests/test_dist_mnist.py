@@ -737,3 +737,49 @@ def _elastic_once(free_ports, d):
     ck = latest_checkpoint(ckdir)
     assert ck is not None
     assert int(ck.rsplit("-", 1)[1]) >= 30
+
+
+def test_async_tcp_transport(monkeypatch):
+    """Cross-host async data path: with ROCPS_ASYNC_TCP=1
+    the worker never maps the shard's /dev/shm segment — every step is
+    one push_pull TCP round trip (push + apply + fresh-param pull,
+    the reference's RunGraph exchange).  Training must converge and
+    the PS counters must match the shm path's semantics."""
+    import torch
+
+    from rocps.cluster import ControlClient
+    from rocps.engine.async_engine import AsyncPSShard, AsyncWorkerEngine
+    from rocps.models import mnist_mlp, synthetic_mnist_batch
+    from rocps.testing import create_local_cluster
+    from rocps.train.optimizer import GradientDescentOptimizer
+
+    monkeypatch.setenv("ROCPS_ASYNC_TCP", "1")
+    workers, ps, cluster = create_local_cluster(num_workers=1, num_ps=2)
+    try:
+        shards = [AsyncPSShard.serve_on(p, cluster) for p in ps]
+        torch.manual_seed(5)
+        m = mnist_mlp(hidden=16)
+        eng = AsyncWorkerEngine(m, cluster,
+                                ControlClient(cluster.master_address), 0,
+                                torch.device("cpu"))
+        opt = GradientDescentOptimizer(0.5)
+        eng.initialize(opt)
+        assert len(eng._tcp) == 2        # both shards went TCP
+        assert all(s is None for s in eng.segs)
+        g = torch.Generator().manual_seed(1)
+        x, y = synthetic_mnist_batch(64, generator=g)
+        losses = []
+        for _ in range(25):
+            eng.zero_grad()
+            loss = m.loss(x, y)
+            eng.step(loss, opt)
+            losses.append(float(loss.detach()))
+        assert losses[-1] < losses[0]
+        assert eng.last_global_step == 25
+        for sh in shards:
+            assert sh.applied == 25 and sh.total_pushes == 25
+            assert sh.dropped_stale == 0
+        eng.detach()
+    finally:
+        for s in workers + ps:
+            s.stop()
