@@ -59,7 +59,11 @@ __device__ __forceinline__ int map_row(int h, int pr) {
   }
 }
 
-template <int EPI>  // 0 none, 1 +bias, 2 +bias+gelu, 3 = 2 + store pre
+template <int EPI, bool INPH = false>
+// EPI: 0 none, 1 +bias, 2 +bias+gelu, 3 = 2 + store pre.
+// INPH: phase reads its OWN MFMA operands (the guide template's
+// literal shape — reads gated by the phase's barrier + the compiler's
+// counted lgkm waits) instead of reading one phase ahead.
 __global__ __launch_bounds__(512)
 void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
                     const float* __restrict__ bias, bf16* __restrict__ D,
@@ -128,6 +132,47 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
 #define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
 #define BAR() __builtin_amdgcn_s_barrier()
 
+  if constexpr (INPH) {
+    // ---- in-phase schedule -----------------------------------------
+    // phase X.P0: read a0(X)+b0(X) [12], MFMA q00, stage A0(X+1)
+    //       X.P1: read b1(X)       [4],  MFMA q01, stage B0(X+1)
+    //       X.P2: read a1(X)       [8],  MFMA q10, stage B1(X+1)
+    //       X.P3: (cached)         [0],  MFMA q11, stage A1(X+1)
+    // vmcnt(4) at P0/P1/P3 guarantees the halves the NEXT phase reads
+    // (staged 3-4 phases earlier); reads are covered by the previous
+    // phase's vmcnt + barrier, and hipcc's counted lgkm waits order
+    // each phase's reads before its own MFMAs.
+#define RD_A(dst, buf, h)                                                  {                                                                          const char* p0 = smem + (buf) * BUF + (h) * HALF + abase;                const char* p1 = smem + (((buf) * BUF + (h) * HALF + abase) ^ 64);       _Pragma("unroll") for (int f = 0; f < 4; f++) {                            dst[f][0] = *(const bf16x8*)(p0 + f * 2048);                             dst[f][1] = *(const bf16x8*)(p1 + f * 2048);                           }                                                                      }
+#define RD_B(dst, buf, h)                                                  {                                                                          const char* p0 = smem + (buf) * BUF + (h) * HALF + bbase;                const char* p1 = smem + (((buf) * BUF + (h) * HALF + bbase) ^ 64);       _Pragma("unroll") for (int f = 0; f < 2; f++) {                            dst[f][0] = *(const bf16x8*)(p0 + f * 2048);                             dst[f][1] = *(const bf16x8*)(p1 + f * 2048);                           }                                                                      }
+#define MM16(am, bm, mb, nb)                                               __builtin_amdgcn_s_setprio(1);                                           _Pragma("unroll") for (int i = 0; i < 4; i++)                                _Pragma("unroll") for (int j = 0; j < 2; j++)                                _Pragma("unroll") for (int kk = 0; kk < 2; kk++)                             acc[(mb) + i][(nb) + j] =                                                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(                                     am[i][kk], bm[j][kk], acc[(mb) + i][(nb) + j],                           0, 0, 0);                                            __builtin_amdgcn_s_setprio(0)
+#define VM4 asm volatile("s_waitcnt vmcnt(4)" ::: "memory")
+    stage(0, H_A0, 0); stage(0, H_B0, 0); stage(0, H_B1, 0);
+    stage(0, H_A1, 0);
+    VM4;  // A0(0), B0(0) landed
+    BAR();
+    for (int t = 0; t < nkt; t += 2) {
+      // tile t (buffer 0)
+      RD_A(a0, 0, H_A0); RD_B(b0, 0, H_B0);
+      stage(1, H_A0, t + 1); VM4; MM16(a0, b0, 0, 0); BAR();  // P0 q00
+      RD_B(b1, 0, H_B1);
+      stage(1, H_B0, t + 1); VM4; MM16(a0, b1, 0, 2); BAR();  // P1 q01
+      RD_A(a1, 0, H_A1);
+      stage(1, H_B1, t + 1);      MM16(a1, b0, 4, 0); BAR();  // P2 q10
+      stage(1, H_A1, t + 1); VM4; MM16(a1, b1, 4, 2); BAR();  // P3 q11
+      // tile t+1 (buffer 1)
+      RD_A(a0, 1, H_A0); RD_B(b0, 1, H_B0);
+      stage(0, H_A0, t + 2); VM4; MM16(a0, b0, 0, 0); BAR();
+      RD_B(b1, 1, H_B1);
+      stage(0, H_B0, t + 2); VM4; MM16(a0, b1, 0, 2); BAR();
+      RD_A(a1, 1, H_A1);
+      stage(0, H_B1, t + 2);      MM16(a1, b0, 4, 0); BAR();
+      stage(0, H_A1, t + 2); VM4; MM16(a1, b1, 4, 2); BAR();
+    }
+#undef RD_A
+#undef RD_B
+#undef MM16
+#undef VM4
+  } else {
   // ---- prologue: 6 halves (tiles 0 and 1), first reads --------------
   stage(0, H_A0, 0); stage(0, H_B0, 0); stage(0, H_B1, 0); stage(0, H_A1, 0);
   stage(1, H_A0, 1); stage(1, H_B0, 1);
@@ -218,6 +263,7 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
 #undef READ_B
 #undef MFMA16
 #undef PHASE_TAIL
+  }
 #undef VMCNT
 #undef BAR
 
@@ -248,12 +294,17 @@ bool gemm256_supported(int M, int N, int K) {
 
 void launch_gemm_bf16_256(const rbf16* A, const rbf16* W, const float* bias,
                           rbf16* D, rbf16* PRE, int M, int N, int K,
-                          int epilogue, hipStream_t st) {
+                          int epilogue, hipStream_t st, int sched) {
   dim3 grid((M / BM) * (N / BN));
   const bf16* a = reinterpret_cast<const bf16*>(A);
   const bf16* w = reinterpret_cast<const bf16*>(W);
   bf16* d = reinterpret_cast<bf16*>(D);
   bf16* p = reinterpret_cast<bf16*>(PRE);
+  if (sched == 1) {  // in-phase schedule (A/B benching)
+    hipLaunchKernelGGL((gemm256_kernel<0, true>), grid, dim3(512), 0, st, a,
+                       w, bias, d, nullptr, M, N, K);
+    return;
+  }
   if (epilogue == 3)
     hipLaunchKernelGGL((gemm256_kernel<3>), grid, dim3(512), 0, st, a, w,
                        bias, d, p, M, N, K);

@@ -118,7 +118,7 @@ void launch_gelu_bwd(const rbf16* pre, const rbf16* dy, rbf16* g, long n,
 bool gemm256_supported(int M, int N, int K);
 void launch_gemm_bf16_256(const rbf16* A, const rbf16* W, const float* bias,
                           rbf16* D, rbf16* PRE, int M, int N, int K,
-                          int epilogue, hipStream_t st);
+                          int epilogue, hipStream_t st, int sched = 0);
 
 // gemm_bf16.hip
 void launch_gemm_bf16(const rbf16* A, const rbf16* W, const float* bias,

@@ -138,11 +138,14 @@ def test_fake_bn_buffer_sync():
     assert torch.equal(after0, after1)  # replicas re-synced in place
 
 
-def test_fake_rs_ag_checkpoint_save_restore_sharded_state():
+@pytest.mark.parametrize("world", [2, 8])
+def test_fake_rs_ag_checkpoint_save_restore_sharded_state(world):
     """Trainer-level save/restore over the rs_ag engine: collecting the
     checkpoint all-gathers each rank's optimizer-state shard and restore
     re-scatters it — the exact path the RCCL engine takes on GPU.
-    12 straight steps == 6 steps + save + restore-into-fresh-world + 6."""
+    12 straight steps == 6 steps + save + restore-into-fresh-world + 6.
+    world=8 is the MI355X node shape (maximize 8-rank
+    readiness evidence without an 8-GPU node)."""
     from rocps.train.session import Trainer
 
     steps, B, lr, mu = 6, 8, 0.1, 0.9
@@ -160,7 +163,7 @@ def test_fake_rs_ag_checkpoint_save_restore_sharded_state():
             tr.engine.broadcast_parameters(src=0)
             for s in range(tr.global_step, tr.global_step + steps):
                 g = torch.Generator().manual_seed(3 + s)
-                x, y = synthetic_mnist_batch(2 * B, generator=g)
+                x, y = synthetic_mnist_batch(world * B, generator=g)
                 lo = rank * B
                 xb, yb = x[lo:lo + B], y[lo:lo + B]
                 tr.train_step(lambda: tr.model.loss(xb, yb))
@@ -169,21 +172,21 @@ def test_fake_rs_ag_checkpoint_save_restore_sharded_state():
                 results[rank] = st
             return rank
 
-        run_fake_cluster(2, rank_fn)
+        run_fake_cluster(world, rank_fn)
         if out_state is not None:
             out_state.update(results[0])
         return results
 
     # straight 12-step reference
     models = []
-    for _ in range(2):
+    for _ in range(world):
         torch.manual_seed(7)
         models.append(mnist_mlp(hidden=32))
     straight = trainer_run(models, 12, 0)
 
     # 6 steps, checkpoint, restore into a FRESH world, 6 more
     models = []
-    for _ in range(2):
+    for _ in range(world):
         torch.manual_seed(7)
         models.append(mnist_mlp(hidden=32))
     saved = {}
@@ -192,7 +195,7 @@ def test_fake_rs_ag_checkpoint_save_restore_sharded_state():
     assert any(k.startswith("optimizer/bucket") for k in saved)
 
     models = []
-    for _ in range(2):
+    for _ in range(world):
         torch.manual_seed(99)  # different init; restore must overwrite
         models.append(mnist_mlp(hidden=32))
     resumed = trainer_run(models, steps, 6, state=saved)
