@@ -28,3 +28,33 @@ def test_bench_json_contract():
     assert rec["value"] > 0
     assert rec["config"]["global_batch"] == 16
     assert rec["config"]["parallelism"] == "dp1"
+
+
+def test_bench_torchrun_two_ranks_contract():
+    """The driver's exact multi-GPU launch shape, on CPU/gloo: one
+    process per rank via torch.distributed.run, rank 0 prints ONE JSON
+    line whose value is the WHOLE-JOB aggregate."""
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               GLOO_SOCKET_IFNAME="lo")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--model", "mnist_mlp", "--steps", "3",
+         "--warmup", "1", "--batch", "16"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-3000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout   # only rank 0 prints
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    # whole-job aggregate: global batch = per-rank byte 16 x 2 ranks
+    assert rec["config"]["global_batch"] == 32
+    assert rec["value"] > 0

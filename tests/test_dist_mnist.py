@@ -783,3 +783,17 @@ def test_async_tcp_transport(monkeypatch):
     finally:
         for s in workers + ps:
             s.stop()
+
+
+def test_sync_four_workers_parity(free_ports, tmp_path):
+    """4-worker sync parity vs the single-process full-batch run —
+    half of the 8-rank node shape, on gloo subprocesses (VERDICT r1
+    #4: multi-GPU readiness evidence on CPU)."""
+    steps = 12
+    multi, _ = _run_cluster(free_ports, tmp_path / "multi", issync=True,
+                            n_workers=4, steps=steps, batch=8)
+    single = _run_single(free_ports, tmp_path / "single", steps=steps,
+                         batch=32)
+    for s in range(steps):
+        avg = sum(multi[i][s] for i in range(4)) / 4
+        assert abs(avg - single[s]) < 1e-4, (s, avg, single[s])
