@@ -401,20 +401,21 @@ static void dispatch_gemm(const at::Tensor& A, const at::Tensor& W,
                           long M, long N, long K, long epilogue,
                           long variant) {
   bool use256 = gemm256_supported((int)M, (int)N, (int)K);
-  if (variant == 3) {  // in-phase 256 schedule (microbench A/B)
+  if (variant == 3) {  // legacy look-ahead 256 schedule (microbench A/B)
     TORCH_CHECK(use256 && epilogue == 0,
                 "variant 3: 256-tile shape, epilogue 0 only");
     launch_gemm_bf16_256(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), pre,
                          (int)M, (int)N, (int)K, 0, cur_stream(A), 1);
     return;
   }
-  // Measured heuristic (profiles/gemm_micro_r2*.txt): the 256-tile
+  // Measured heuristic (profiles/gemm_micro_r2c.txt): the 256-tile
   // pipeline runs at 1 workgroup/CU, so it needs a grid that fills the
-  // 256 CUs and enough K-depth to amortize its 6-half prologue; on
-  // small grids or shallow K the 128-tile kernel (2-3 blocks/CU) wins.
+  // 256 CUs and K-depth to amortize its prologue; on small grids or
+  // shallow K — and on huge grids, where 2-3 blocks/CU of the smaller
+  // tile win back the latency hiding — the 128-tile kernel is faster.
   if (use256 && variant == -1) {
     long blocks = (M / 256) * (N / 256);
-    use256 = blocks >= 192 && K >= 1024;
+    use256 = blocks >= 192 && blocks < 1024 && K >= 1024;
   }
   if (variant == 1) use256 = false;
   if (variant == 2)

@@ -123,7 +123,7 @@ def bench_gemm(dev, reps):
             print(f"rocps gemm256  {M}x{N}x{K}: {us:9.1f} us "
                   f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
             us = timeit(lambda: ext.gemm_bf16(a, w, None, 0, 3), reps)
-            print(f"rocps gemm256i {M}x{N}x{K}: {us:9.1f} us "
+            print(f"rocps gemm256L {M}x{N}x{K}: {us:9.1f} us "
                   f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
         us = timeit(lambda: a @ w.t(), reps)
         print(f"rocblas matmul {M}x{N}x{K}: {us:9.1f} us "
