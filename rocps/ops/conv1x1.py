@@ -86,9 +86,10 @@ def conv1x1_nhwc(x: torch.Tensor, weight: torch.Tensor,
 # dgrad   : dx_rows  = dy_rows(M, Cout) @ (W^T)(Cin, Cout)^T   [MFMA GEMM]
 #           (the weight transpose is a (Cout x Cin) copy — trivially small
 #           next to the activation tensors)
-# wgrad   : dW = dy_rows^T @ rows                              [rocBLAS TN]
-#           (contraction over M with both operands M-major; a plain
-#           library GEMM per the MI355X design rules)
+# wgrad   : MIOpen's wrw asm kernel on GPU (convolution_backward with
+#           the weight-only output mask — the library kernel for the
+#           tall-skinny sum over M, exactly like rocBLAS for plain
+#           GEMMs); rocBLAS/fp32 TN on CPU
 #
 # Dispatch is PER SHAPE: only (cin, cout, stride) combinations measured
 # faster than MIOpen's igemm asm (tools/bench_conv1x1.py, fwd+bwd) run
@@ -149,17 +150,25 @@ class _Conv1x1Fn(torch.autograd.Function):
         s = ctx.stride
         n, cin, h, wd = ctx.in_shape
         cout = wmat.shape[0]
-        dyr = (dy.contiguous(memory_format=torch.channels_last)
-               .permute(0, 2, 3, 1).reshape(-1, cout))
-        # dgrad: dx_rows[m, cin] = sum_cout dy[m, cout] * W[cout, cin]
-        dxr = _nt(dyr, wmat.t().contiguous())
-        # wgrad: contraction over the huge M dim — library GEMM
-        if rows.is_cuda:
-            dw = dyr.t() @ rows
-        else:
-            dw = (dyr.float().t() @ rows.float()).to(wmat.dtype)
+        dyc = dy.contiguous(memory_format=torch.channels_last)
+        dyr = dyc.permute(0, 2, 3, 1).reshape(-1, cout)
         ho = (h + s - 1) // s
         wo = (wd + s - 1) // s
+        # dgrad: dx_rows[m, cin] = sum_cout dy[m, cout] * W[cout, cin]
+        dxr = _nt(dyr, wmat.t().contiguous())
+        if rows.is_cuda:
+            # wgrad over the huge M contraction: MIOpen's wrw kernel
+            # (the fwd ran on the stride-1 subsampled xs, so wrw of a
+            # stride-1 1x1 conv over xs is exactly dW)
+            xs = (rows.reshape(n, ho, wo, cin).permute(0, 3, 1, 2)
+                  .contiguous(memory_format=torch.channels_last))
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dyc, xs, wmat.reshape(cout, cin, 1, 1), None,
+                [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, False])
+            dw = dw.reshape(cout, cin)
+        else:
+            dw = (dyr.float().t() @ rows.float()).to(wmat.dtype)
         dxs = (dxr.reshape(n, ho, wo, cin).permute(0, 3, 1, 2)
                .contiguous(memory_format=torch.channels_last))
         if s > 1:
