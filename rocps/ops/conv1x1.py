@@ -102,11 +102,16 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 # (cin, cout, stride) -> GEMM path wins fwd+bwd vs MIOpen on MI355X at
-# batch 256 (profiles/conv1x1_train_r2.txt): the 7x7-stage projections,
-# where the small spatial M keeps the rocBLAS TN wgrad cheap.  On the
-# large-M shapes MIOpen's dedicated wrw asm kernels win the backward;
-# a hand-written split-M wgrad kernel is the open item.
-_GEMM_SHAPES = {(512, 2048, 1), (2048, 512, 1)}
+# batch 256 (profiles/conv1x1_train_r2b.txt): with MIOpen's wrw kernel
+# carrying the weight gradient, the MFMA fwd+dgrad wins 10 of 15
+# ResNet-50 1x1 shapes (-10..-17% per conv fwd+bwd).  Still MIOpen:
+# the stride-2 downsamples (subsample+scatter overhead) and the
+# Cin=64 stem-adjacent shapes.
+_GEMM_SHAPES = {
+    (128, 512, 1), (256, 64, 1), (256, 128, 1), (256, 1024, 1),
+    (512, 128, 1), (512, 256, 1), (512, 2048, 1), (1024, 256, 1),
+    (1024, 512, 1), (2048, 512, 1),
+}
 
 
 def set_gemm_shapes(shapes) -> None:
