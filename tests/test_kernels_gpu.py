@@ -444,11 +444,10 @@ def test_maxpool332_bf16_forward_exact():
     assert torch.equal(y, ref)
 
 
-@pytest.mark.skipif(os.environ.get("ROCPS_TEST_STAGED") != "1",
-                    reason="staged round-2 kernel (set ROCPS_TEST_STAGED=1)")
-def test_gemm_bf16_gelu_pre_staged():
-    """Epilogue-3 (gelu + pre-activation out) — staged for round 2's
-    BERT training path; run with ROCPS_TEST_STAGED=1."""
+def test_gemm_bf16_gelu_pre():
+    """Epilogue-3 (gelu + pre-activation out) — the BERT training
+    MLP's fused forward (un-gated since round 2 wired it into
+    BertLayer._mlp under grad)."""
     from rocps.ops import hip_ext
 
     g = torch.Generator().manual_seed(0)
