@@ -88,7 +88,6 @@ class ElasticRecovery:
         c.put(f"regroup/{epoch}/task{t.task_index}", 1)
         expected = sorted(self.members)
         deadline = time.time() + self.join_cap
-        quiesce = time.time() + self.join_window
         alive = {t.task_index}
         while True:
             dead = self._dead()
@@ -100,15 +99,16 @@ class ElasticRecovery:
                     alive.add(i)
                 except KeyError:
                     pass
-            now = time.time()
+            # exit when every member is accounted for: posted (alive)
+            # or heartbeat-declared dead.  A live straggler keeps
+            # heartbeating from its daemon thread even while its main
+            # thread is stuck in a failing collective, so it is waited
+            # for rather than dropped.
             if all(i in alive or i in dead for i in expected):
                 break
-            if now > deadline:
+            if time.time() > deadline:
                 break
             time.sleep(0.2)
-            # settle at least join_window so stragglers make the room
-            if now < quiesce:
-                continue
         return sorted(alive)
 
     # ------------------------------------------------------------------
