@@ -105,7 +105,7 @@ def main(_):
             from rocps.train.recovery import ElasticRecovery
             recovery = ElasticRecovery(
                 trainer, FLAGS.checkpoint_dir or None,
-                join_window=6.0, comm_timeout=30.0, dead_threshold=5.0)
+                comm_timeout=30.0, dead_threshold=5.0)
         local_step = 0
         while not sess.should_stop():
             # deterministic synthetic batch for this global step; each

@@ -40,11 +40,10 @@ class WorkerLost(RuntimeError):
 
 class ElasticRecovery:
     def __init__(self, trainer, checkpoint_dir: Optional[str],
-                 join_window: float = 8.0, comm_timeout: float = 60.0,
+                 comm_timeout: float = 60.0,
                  dead_threshold: float = 6.0, join_cap: float = 45.0):
         self.trainer = trainer
         self.checkpoint_dir = checkpoint_dir
-        self.join_window = join_window
         self.comm_timeout = comm_timeout
         self.dead_threshold = dead_threshold
         self.join_cap = join_cap
