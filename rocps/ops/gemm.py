@@ -1,15 +1,23 @@
 """Hand-written MFMA bf16 GEMM — capability C10.
 
 ``matmul_nt(a, w)`` computes a @ w.T (the torch Linear contract) on the
-CDNA4 MFMA kernel (csrc/kernels/gemm_bf16.hip: 128x128 macro-tile,
-global_load_lds staging, XOR-swizzled LDS, 16x16x32 bf16 MFMA), with
-optional fused bias and tanh-GELU epilogues — the SURVEY C10 "bias+GELU
-epilogue variant for BERT".
+CDNA4 MFMA kernels, with optional fused bias and tanh-GELU epilogues —
+the SURVEY C10 "bias+GELU epilogue variant for BERT".  Two kernels sit
+behind one dispatch (csrc/kernels/gemm_bf16.hip / gemm_bf16_256.hip):
 
-Plain (unfused) GEMMs in the models run through rocBLAS/hipBLASLt via
-torch.nn.Linear, per the MI355X design rules; this kernel is the fused
-forward building block and the shape contract is M%128 == N%128 == 0,
-K%64 == 0 (use ``supported(...)`` to check).
+- 128x128 macro-tile, global_load_lds staging, XOR-swizzled LDS,
+  16x16x32 bf16 MFMA — 2-3 blocks/CU, wins small grids, shallow K and
+  huge grids;
+- 256x256 8-phase pipeline (8 waves, one half-tile staged per phase,
+  counted s_waitcnt vmcnt across raw barriers, s_setprio'd 16-MFMA
+  clusters) — 1 block/CU, wins mid grids at K>=1024 (1.12 PF/s at
+  4096^3 vs 0.86 for the 128-tile kernel, profiles/gemm_micro_r2c.txt).
+
+They run on the training hot path: conv1x1 forward+dgrad in
+ResNet/W-ResNet (rocps/ops/conv1x1.py) and the BERT MLP fc1 (epilogue
+3 emits the pre-activation for backward reuse).  Plain library GEMMs
+elsewhere go through rocBLAS via torch.nn.Linear.  Shape contract
+M%128 == N%128 == 0, K%64 == 0 (``supported(...)``).
 """
 
 from __future__ import annotations
