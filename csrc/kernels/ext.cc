@@ -401,11 +401,13 @@ static void dispatch_gemm(const at::Tensor& A, const at::Tensor& W,
                           long M, long N, long K, long epilogue,
                           long variant) {
   bool use256 = gemm256_supported((int)M, (int)N, (int)K);
-  if (variant == 3) {  // legacy look-ahead 256 schedule (microbench A/B)
+  if (variant == 3 || variant == 4) {  // A/B: 3 = look-ahead schedule,
+    // 4 = in-phase with the st_16x32 swizzle
     TORCH_CHECK(use256 && epilogue == 0,
-                "variant 3: 256-tile shape, epilogue 0 only");
+                "variant 3/4: 256-tile shape, epilogue 0 only");
     launch_gemm_bf16_256(bf16_ptr(A), bf16_ptr(W), bptr, bf16_ptr(D), pre,
-                         (int)M, (int)N, (int)K, 0, cur_stream(A), 1);
+                         (int)M, (int)N, (int)K, 0, cur_stream(A),
+                         variant == 3 ? 1 : 2);
     return;
   }
   // Measured heuristic (profiles/gemm_micro_r2c.txt): the 256-tile

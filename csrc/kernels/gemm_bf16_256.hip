@@ -59,11 +59,14 @@ __device__ __forceinline__ int map_row(int h, int pr) {
   }
 }
 
-template <int EPI, bool INPH = false>
+template <int EPI, bool INPH = false, int SWZ = 0>
 // EPI: 0 none, 1 +bias, 2 +bias+gelu, 3 = 2 + store pre.
 // INPH: phase reads its OWN MFMA operands (the guide template's
 // literal shape — reads gated by the phase's barrier + the compiler's
 // counted lgkm waits) instead of reading one phase ahead.
+// SWZ: LDS bank swizzle — 0: byte ^= (pr&7)<<4 (spread over 8 16-B
+// slots); 1: the guide's st_16x32 (byte bit5 ^= pr bit2, 4-way
+// spread, and the kk step stays a plain +64 immediate).
 __global__ __launch_bounds__(512)
 void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
                     const float* __restrict__ bias, bf16* __restrict__ D,
@@ -101,7 +104,7 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
     for (int i = 0; i < 2; i++) {
       int off = i * 8192 + tid * 16;
       int pr = off >> 7, b = off & 127;
-      int sb = b ^ ((pr & 7) << 4);
+      int sb = SWZ ? (b ^ ((pr & 4) << 3)) : (b ^ ((pr & 7) << 4));
       const bf16* src = base + (long)map_row(h, pr) * K + kt * BK + (sb >> 1);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
@@ -117,7 +120,7 @@ void gemm256_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   // bits 4-6, the row field above bit 7 — no carries), so it costs one
   // v_xor instead of a second live-range (a spilled address VGPR here
   // forces a compiler vmcnt(0) that drains the whole glds pipeline).
-  const int xorv = (lane & 7) << 4;
+  const int xorv = SWZ ? ((lane & 4) << 3) : ((lane & 7) << 4);
   const int rdb = (((lane >> 4) << 4) ^ xorv);
   // a-frags: pr = wr*64 + f*16 + (lane&15)   (f = 0..3)
   // b-frags: pr = wc*32 + j*16 + (lane&15)   (j = 0..1 within the half)
@@ -302,6 +305,11 @@ void launch_gemm_bf16_256(const rbf16* A, const rbf16* W, const float* bias,
   bf16* p = reinterpret_cast<bf16*>(PRE);
   if (sched == 1) {  // legacy look-ahead schedule (A/B benching)
     hipLaunchKernelGGL((gemm256_kernel<0, false>), grid, dim3(512), 0, st,
+                       a, w, bias, d, nullptr, M, N, K);
+    return;
+  }
+  if (sched == 2) {  // in-phase + st_16x32 swizzle (A/B benching)
+    hipLaunchKernelGGL((gemm256_kernel<0, true, 1>), grid, dim3(512), 0, st,
                        a, w, bias, d, nullptr, M, N, K);
     return;
   }

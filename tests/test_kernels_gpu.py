@@ -478,12 +478,15 @@ def test_gemm256_vs_128_bitwise(shape):
     w = (torch.randn(N, K, device=DEV) / K ** 0.25).bfloat16()
     d256 = hip_ext().gemm_bf16(a, w, None, 0, 2)     # in-phase schedule
     d256i = hip_ext().gemm_bf16(a, w, None, 0, 3)    # look-ahead schedule
+    d256s = hip_ext().gemm_bf16(a, w, None, 0, 4)    # st_16x32 swizzle
     d128 = hip_ext().gemm_bf16(a, w, None, 0, 1)
     torch.cuda.synchronize()
     assert torch.equal(d256, d128), (
         (d256 != d128).sum().item(), "mismatched elements")
     assert torch.equal(d256i, d128), (
         (d256i != d128).sum().item(), "look-ahead schedule mismatch")
+    assert torch.equal(d256s, d128), (
+        (d256s != d128).sum().item(), "st_16x32 swizzle mismatch")
 
 
 def test_gemm256_numerics_vs_fp32():

@@ -125,6 +125,9 @@ def bench_gemm(dev, reps):
             us = timeit(lambda: ext.gemm_bf16(a, w, None, 0, 3), reps)
             print(f"rocps gemm256L {M}x{N}x{K}: {us:9.1f} us "
                   f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
+            us = timeit(lambda: ext.gemm_bf16(a, w, None, 0, 4), reps)
+            print(f"rocps gemm256S {M}x{N}x{K}: {us:9.1f} us "
+                  f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
         us = timeit(lambda: a @ w.t(), reps)
         print(f"rocblas matmul {M}x{N}x{K}: {us:9.1f} us "
               f"{flop / (us * 1e-6) / 1e12:7.1f} TF/s", flush=True)
