@@ -559,3 +559,28 @@ def test_gelu_bwd_kernel():
         torch.cuda.synchronize()
         ref = dy.float() * _dgelu_tanh(pre.float())
         assert torch.allclose(got.float(), ref, atol=2e-2, rtol=3e-2), n
+
+
+def test_resnet_bottleneck_uses_conv1x1_gemm_dispatch():
+    """Dispatch guard: inside a real Bottleneck on GPU bf16, 1x1 convs
+    whose (cin, cout, stride) is in the measured win table must run
+    the MFMA-GEMM autograd Function — a silent fallback to MIOpen
+    would pass numerics tests while losing the hand-written path."""
+    from rocps.models.resnet import Bottleneck
+    from rocps.ops.conv1x1 import _GEMM_SHAPES
+
+    torch.manual_seed(3)
+    # layer3-style block: 1024 -> 256 -> 1024 (both 1x1s in the table)
+    blk = (Bottleneck(1024, 256).to(DEV).bfloat16()
+           .to(memory_format=torch.channels_last))
+    assert (1024, 256, 1) in _GEMM_SHAPES
+    assert (256, 1024, 1) in _GEMM_SHAPES
+    x = (torch.randn(4, 1024, 14, 14, device=DEV).bfloat16()
+         .contiguous(memory_format=torch.channels_last).requires_grad_(True))
+    out = blk.conv1(x)
+    assert type(out.grad_fn).__name__.startswith("_Conv1x1Fn"), \
+        type(out.grad_fn).__name__
+    # and the full block trains through it
+    y = blk(x)
+    y.float().square().mean().backward()
+    assert x.grad is not None and torch.isfinite(x.grad.float()).all()
