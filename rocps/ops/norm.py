@@ -162,11 +162,30 @@ class BatchNormAct2d(nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
+        # counted host-side: the buffer lives on GPU with the model and
+        # a per-forward `+= 1` is one tiny kernel launch per BN layer
+        # per step (53/step on ResNet-50); the python counter folds
+        # into the buffer at state_dict/checkpoint time
+        self._nbt_pending = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self._flush_nbt()
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+
+    def _flush_nbt(self) -> None:
+        if self._nbt_pending:
+            with torch.no_grad():
+                self.num_batches_tracked += self._nbt_pending
+            self._nbt_pending = 0
+
+    def _load_from_state_dict(self, *args, **kw):
+        self._nbt_pending = 0
+        super()._load_from_state_dict(*args, **kw)
 
     def forward(self, x: torch.Tensor,
                 residual: Optional[torch.Tensor] = None) -> torch.Tensor:
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt_pending += 1
         if self.act == "add_relu":
             assert residual is not None, "add_relu needs a residual input"
         return fused_bn_act(x, self.weight, self.bias, self.running_mean,

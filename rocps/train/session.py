@@ -151,6 +151,11 @@ class Trainer:
             # them across replicas (the reference's PS-shared moving
             # stats), so the checkpoint sees cluster-wide statistics
             self.engine.sync_buffers()
+        # lazily-counted stats (BatchNormAct2d num_batches_tracked)
+        # fold into their buffers before collection
+        for mod in self.model.modules():
+            if hasattr(mod, "_flush_nbt"):
+                mod._flush_nbt()
         out: Dict[str, torch.Tensor] = {
             name: p.detach() for name, p in self.model.named_parameters()}
         for name, b in self.model.named_buffers():

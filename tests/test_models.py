@@ -94,3 +94,24 @@ def test_bn_eval_mode_backward_matches_frozen_stats():
         assert torch.allclose(x.grad, x2.grad, atol=1e-5), (relu,)
         assert torch.allclose(w.grad, w2.grad, atol=1e-4), (relu,)
         assert torch.allclose(b.grad, b2.grad, atol=1e-4), (relu,)
+
+
+def test_bn_num_batches_tracked_lazy_counter():
+    """num_batches_tracked counts host-side per forward and folds into
+    the buffer at state_dict time (no per-layer GPU launch per step)."""
+    from rocps.ops.norm import BatchNormAct2d
+    m = BatchNormAct2d(4, act="none")
+    x = torch.randn(2, 4, 3, 3)
+    for _ in range(5):
+        m(x)
+    assert m._nbt_pending == 5
+    sd = m.state_dict()
+    assert int(sd["num_batches_tracked"]) == 5
+    assert m._nbt_pending == 0
+    m(x)
+    sd = m.state_dict()
+    assert int(sd["num_batches_tracked"]) == 6
+    # load resets the pending counter
+    m2 = BatchNormAct2d(4, act="none")
+    m2.load_state_dict(sd)
+    assert int(m2.num_batches_tracked) == 6 and m2._nbt_pending == 0
