@@ -110,13 +110,30 @@ def main(_):
             save_checkpoint_steps=(FLAGS.save_checkpoint_steps or None),
             save_checkpoint_secs=(None if FLAGS.save_checkpoint_steps
                                   else 600)) as sess:
-        while not sess.should_stop():
-            x, y = make_batch(FLAGS.model, FLAGS.batch_size, FLAGS.seq_len,
-                              trainer.global_step, FLAGS.seed,
-                              trainer.device)
-            loss = sess.run(lambda: trainer.model.loss(x, y))
-            if events is not None and trainer.global_step % 10 == 0:
-                events.add_scalar("loss", float(loss), trainer.global_step)
+        # input feeding (C14): producer threads synthesize the
+        # step-keyed batches into a bounded pinned queue (the
+        # reference's queue runners); the consumer's H2D copy is then
+        # a pinned DMA that overlaps compute.  Seeds stay exact across
+        # restores: producer index i feeds step base + i.
+        base = trainer.global_step
+        use_cuda = trainer.device.type == "cuda"
+        qr = rocps.train.QueueRunner(
+            lambda i: make_batch(FLAGS.model, FLAGS.batch_size,
+                                 FLAGS.seq_len, base + i, FLAGS.seed,
+                                 torch.device("cpu")),
+            capacity=4, num_threads=2, pin_memory=use_cuda)
+        qcoord = rocps.train.start_queue_runners([qr])
+        try:
+            while not sess.should_stop():
+                x, y = qr.dequeue()
+                x = x.to(trainer.device, non_blocking=True)
+                y = y.to(trainer.device, non_blocking=True)
+                loss = sess.run(lambda: trainer.model.loss(x, y))
+                if events is not None and trainer.global_step % 10 == 0:
+                    events.add_scalar("loss", float(loss),
+                                      trainer.global_step)
+        finally:
+            qcoord.request_stop()
     if events is not None:
         events.close()
 

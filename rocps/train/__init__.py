@@ -22,6 +22,8 @@ from rocps.train.saver import (  # noqa: F401
 from rocps.train.coordinator import Coordinator, LooseVersionThread  # noqa: F401
 from rocps.train.global_step import (  # noqa: F401
     get_or_create_global_step, get_global_step)
+from rocps.train.queue_runner import (  # noqa: F401
+    QueueRunner, start_queue_runners)
 from rocps.train.recovery import ElasticRecovery, WorkerLost  # noqa: F401
 from rocps.train.supervisor import Supervisor  # noqa: F401
 from rocps.train.schedules import (  # noqa: F401

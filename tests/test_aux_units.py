@@ -333,3 +333,49 @@ def test_optimizer_spec_roundtrip(make):
     for o, p in ((opt, p1), (rebuilt, p2)):
         o.apply_flat(p, g.clone(), slot=0)
     assert torch.equal(p1, p2)
+
+
+def test_queue_runner_prefetch_and_stop():
+    """C14 input feeding: producer threads fill a bounded queue; the
+    consumer sees every index exactly once (per-runner counter);
+    coordinator stop drains cleanly."""
+    import torch
+
+    from rocps.train.coordinator import Coordinator
+    from rocps.train.queue_runner import QueueRunner, start_queue_runners
+
+    def batch_fn(i):
+        g = torch.Generator().manual_seed(i)
+        return i, torch.randn(4, generator=g)
+
+    qr = QueueRunner(batch_fn, capacity=3, num_threads=2)
+    coord = start_queue_runners([qr])
+    seen = set()
+    for _ in range(20):
+        i, t = qr.dequeue(timeout=10)
+        assert i not in seen
+        seen.add(i)
+        g = torch.Generator().manual_seed(i)
+        assert torch.equal(t, torch.randn(4, generator=g))
+    coord.request_stop()
+    coord.join()
+
+
+def test_queue_runner_producer_exception_routes_to_coordinator():
+    import pytest
+
+    from rocps.train.queue_runner import QueueRunner, start_queue_runners
+
+    def bad_fn(i):
+        if i >= 3:
+            raise ValueError("boom")
+        return i
+
+    qr = QueueRunner(bad_fn, capacity=2, num_threads=1)
+    coord = start_queue_runners([qr])
+    got = []
+    with pytest.raises((RuntimeError, TimeoutError)):
+        for _ in range(10):
+            got.append(qr.dequeue(timeout=5))
+    assert coord.should_stop()
+    assert got == [0, 1, 2] or len(got) <= 3
