@@ -457,6 +457,22 @@ at::Tensor gemm_bf16(at::Tensor A, at::Tensor W,
   return D;
 }
 
+at::Tensor conv1x1_wgrad(at::Tensor dY, at::Tensor X) {
+  TORCH_CHECK(dY.is_cuda() && dY.dim() == 2 && dY.is_contiguous());
+  TORCH_CHECK(X.is_cuda() && X.dim() == 2 && X.is_contiguous());
+  TORCH_CHECK(dY.scalar_type() == at::kBFloat16
+              && X.scalar_type() == at::kBFloat16);
+  long M = dY.size(0), N = dY.size(1), K = X.size(1);
+  TORCH_CHECK(X.size(0) == M, "M mismatch");
+  TORCH_CHECK(wgrad_supported(M, (int)N, (int)K),
+              "conv1x1_wgrad needs N%64==0 and K%64==0");
+  c10::hip::OptionalHIPGuardMasqueradingAsCUDA guard(dY.device());
+  auto dW = at::zeros({N, K}, dY.options().dtype(at::kFloat));
+  launch_conv1x1_wgrad(bf16_ptr(dY), bf16_ptr(X), dW.data_ptr<float>(), M,
+                       (int)N, (int)K, cur_stream(dY));
+  return dW;
+}
+
 at::Tensor gelu_bwd(at::Tensor pre, at::Tensor dy) {
   TORCH_CHECK(pre.is_cuda() && pre.scalar_type() == at::kBFloat16);
   TORCH_CHECK(dy.sizes() == pre.sizes()
@@ -557,6 +573,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("variant") = -1);
   m.def("gemm_bf16_gelu_pre", &gemm_bf16_gelu_pre);
   m.def("gelu_bwd", &gelu_bwd);
+  m.def("conv1x1_wgrad", &conv1x1_wgrad);
   m.def("maxpool332_fwd", &maxpool332_fwd);
   m.def("maxpool332_bwd", &maxpool332_bwd);
 }

@@ -110,6 +110,11 @@ void launch_bn_bwd_finalize_bf16(const float* part, int rb, const rbf16* w,
                                  long M, float* abd, rbf16* dgamma,
                                  rbf16* dbeta, int C, hipStream_t st);
 
+// wgrad.hip (conv1x1 weight gradient, tall-skinny reduction GEMM)
+bool wgrad_supported(long M, int N, int K);
+void launch_conv1x1_wgrad(const rbf16* dY, const rbf16* X, float* dW,
+                          long M, int N, int K, hipStream_t st);
+
 // gelu.hip
 void launch_gelu_bwd(const rbf16* pre, const rbf16* dy, rbf16* g, long n,
                      hipStream_t st);

@@ -57,6 +57,7 @@ _hip_sources = [
     "csrc/kernels/ext.cc",
     "csrc/kernels/fused_optimizers.hip",
     "csrc/kernels/softmax_xent.hip",
+    "csrc/kernels/wgrad.hip",
     "csrc/kernels/bn_relu.hip",
     "csrc/kernels/gelu.hip",
     "csrc/kernels/gemm_bf16.hip",

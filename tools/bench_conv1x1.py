@@ -90,4 +90,47 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    if "--wgrad" not in sys.argv:
+        main()
+
+
+def bench_wgrad(batch=256):
+    """A/B the hand-written wgrad kernel vs MIOpen's wrw per shape."""
+    import torch
+    import torch.nn.functional as F
+
+    from rocps.ops import hip_ext
+    ext = hip_ext()
+    print(f"{'shape':>26} {'miopen us':>10} {'custom us':>10} {'win':>5}")
+    for cin, cout, h, s in SHAPES:
+        if s != 1:
+            continue
+        m = batch * h * h
+        x = torch.randn(batch, cin, h, h, device="cuda",
+                        dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        dy = torch.randn(batch, cout, h, h, device="cuda",
+                         dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        w = torch.randn(cout, cin, 1, 1, device="cuda",
+                        dtype=torch.bfloat16)
+        xr = x.permute(0, 2, 3, 1).reshape(m, cin)
+        dyr = dy.permute(0, 2, 3, 1).reshape(m, cout)
+
+        def miopen_wrw():
+            torch.ops.aten.convolution_backward(
+                dy, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, False])
+
+        def custom():
+            ext.conv1x1_wgrad(dyr, xr)
+
+        t_ref = timeit(miopen_wrw)
+        t_cus = timeit(custom)
+        print(f"{(cin, cout, h)!s:>26} {t_ref:10.1f} {t_cus:10.1f} "
+              f"{'CUST' if t_cus < t_ref else 'ref':>5}")
+
+
+if __name__ == "__main__" and "--wgrad" in sys.argv:
+    torch.backends.cudnn.benchmark = True
+    bench_wgrad()
