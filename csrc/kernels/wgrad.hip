@@ -33,32 +33,66 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 constexpr int NT = 64;   // n-tile (dW rows)
 constexpr int KT = 64;   // k-tile (dW cols)
-constexpr int MT = 64;   // m rows staged per iteration
-// one operand tile: 4 panels x [MT][16] bf16 = 8 KB; two operands,
-// double-buffered: 32 KB LDS
+constexpr int MT = 128;  // m rows staged per iteration
+// one operand tile: 4 panels x [MT][16] bf16 = 16 KB; two operands,
+// double-buffered: 64 KB LDS
 constexpr int PANEL = MT * 16;          // elements per panel
 constexpr int TILE = 4 * PANEL;         // elements per operand tile
+constexpr int PASSES = MT * 64 / (256 * 8);  // 16-B moves per thread
 
-__device__ __forceinline__ void stage_tile(
+// T14 split staging: LOAD (issue early, before the MFMA phase hides
+// the HBM latency) and WRITE (after the compute, just before the
+// barrier).  The full-tile path carries no per-load predicate (a
+// runtime select around loads makes hipcc branch + drain per element
+// — guide trap 4c); only the M-tail takes the predicated path.
+template <int PASSES>
+__device__ __forceinline__ void stage_load(
     const bf16* __restrict__ src, long row0, long rows, int ld, int c0,
-    __attribute__((address_space(3))) bf16* dst, int tid) {
-  // 256 threads stage [MT][64c] into 4 panels of [MT][16].  Each
-  // thread moves 16 B (8 c = half a panel row) per pass, 2 passes.
-  // src row r may be out of range (M tail / splitm tail): zero-fill.
+    int tid, bf16x8* v) {
 #pragma unroll
-  for (int pass = 0; pass < 2; pass++) {
-    int idx = tid + pass * 256;        // 512 half-rows-of-panel total
-    int m = idx >> 3;                  // 0..63
-    int half = idx & 7;                // which 8-c chunk of the 64 c
+  for (int pass = 0; pass < PASSES; pass++) {
+    int idx = tid + pass * 256;
+    int m = idx >> 3;
+    int half = idx & 7;
     int panel = half >> 1;
     int off8 = (half & 1) * 8;
-    bf16x8 v = {};
+    v[pass] = *(const bf16x8*)(src + (row0 + m) * (long)ld + c0 +
+                               panel * 16 + off8);
+  }
+}
+
+template <int PASSES>
+__device__ __forceinline__ void stage_load_tail(
+    const bf16* __restrict__ src, long row0, long rows, int ld, int c0,
+    int tid, bf16x8* v) {
+#pragma unroll
+  for (int pass = 0; pass < PASSES; pass++) {
+    int idx = tid + pass * 256;
+    int m = idx >> 3;
+    bf16x8 z = {};
+    v[pass] = z;
     if (m < rows) {
-      v = *(const bf16x8*)(src + (row0 + m) * (long)ld + c0 + panel * 16 +
-                           off8);
+      int half = idx & 7;
+      int panel = half >> 1;
+      int off8 = (half & 1) * 8;
+      v[pass] = *(const bf16x8*)(src + (row0 + m) * (long)ld + c0 +
+                                 panel * 16 + off8);
     }
+  }
+}
+
+template <int PASSES>
+__device__ __forceinline__ void stage_write(
+    __attribute__((address_space(3))) bf16* dst, int tid, const bf16x8* v) {
+#pragma unroll
+  for (int pass = 0; pass < PASSES; pass++) {
+    int idx = tid + pass * 256;
+    int m = idx >> 3;
+    int half = idx & 7;
+    int panel = half >> 1;
+    int off8 = (half & 1) * 8;
     *(__attribute__((address_space(3))) bf16x8*)
-        (dst + panel * PANEL + m * 16 + off8) = v;
+        (dst + panel * PANEL + m * 16 + off8) = v[pass];
   }
 }
 
@@ -110,20 +144,30 @@ void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   f32x4 acc[2][2] = {};
   const int g = lane >> 4;             // 8-m chunk selector
 
+  bf16x8 vy[PASSES], vx[PASSES];
+  auto load_chunk = [&](long r) {
+    long rows = row1 - r;
+    if (rows >= MT) {
+      stage_load<PASSES>(dY, r, rows, N, n0, tid, vy);
+      stage_load<PASSES>(X, r, rows, K, k0, tid, vx);
+    } else {
+      stage_load_tail<PASSES>(dY, r, rows, N, n0, tid, vy);
+      stage_load_tail<PASSES>(X, r, rows, K, k0, tid, vx);
+    }
+  };
+
   int buf = 0;
-  stage_tile(dY, row0, row1 - row0, N, n0, dyt(buf), tid);
-  stage_tile(X, row0, row1 - row0, K, k0, xt(buf), tid);
+  load_chunk(row0);
+  stage_write<PASSES>(dyt(buf), tid, vy);
+  stage_write<PASSES>(xt(buf), tid, vx);
+  if (row0 + MT < row1) load_chunk(row0 + MT);  // issue next loads early
   __syncthreads();
 
   for (long r = row0; r < row1; r += MT) {
-    long rn = r + MT;
-    if (rn < row1) {  // prefetch next MT rows into the other buffer
-      stage_tile(dY, rn, row1 - rn, N, n0, dyt(buf ^ 1), tid);
-      stage_tile(X, rn, row1 - rn, K, k0, xt(buf ^ 1), tid);
-    }
-    // two 32-m MFMA steps over the resident tile
+    // MFMA steps over the resident tile; the NEXT chunk's global
+    // loads are already in flight (issued before the barrier)
 #pragma unroll
-    for (int ms = 0; ms < 2; ms++) {
+    for (int ms = 0; ms < MT / 32; ms++) {
       int mbase = ms * 32 + g * 8;
       bf16x8 a[2], b[2];
 #pragma unroll
@@ -138,8 +182,15 @@ void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
-    buf ^= 1;
+    long rn = r + MT;
+    if (rn < row1) {
+      __syncthreads();  // everyone done reading buf^1's OLD content
+      stage_write<PASSES>(dyt(buf ^ 1), tid, vy);
+      stage_write<PASSES>(xt(buf ^ 1), tid, vx);
+      if (rn + MT < row1) load_chunk(rn + MT);  // re-issue immediately
+      __syncthreads();
+      buf ^= 1;
+    }
   }
 
   // C-map: col = lane&15 (k), row = (lane>>4)*4 + r (n)
