@@ -31,29 +31,30 @@ using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int NT = 64;   // n-tile (dW rows)
-constexpr int KT = 64;   // k-tile (dW cols)
-constexpr int MT = 128;  // m rows staged per iteration
-// one operand tile: 4 panels x [MT][16] bf16 = 16 KB; two operands,
-// double-buffered: 64 KB LDS
-constexpr int PANEL = MT * 16;          // elements per panel
-constexpr int TILE = 4 * PANEL;         // elements per operand tile
-constexpr int PASSES = MT * 64 / (256 * 8);  // 16-B moves per thread
+// Tile geometry is templated: the 64x64 tile (4 panels each op, 128 m
+// rows) wins when N or K is 64; the 128x128 tile (8 panels, 64 m rows
+// — same 64 KB LDS) halves the cross-tile re-read traffic when both
+// dims are >=128.  Waves are fixed 2x2; per-wave fragments = NP/2 x
+// KP/2 of 16x16.
+template <int MT_>
+constexpr int panel_elems() { return MT_ * 16; }
 
 // T14 split staging: LOAD (issue early, before the MFMA phase hides
 // the HBM latency) and WRITE (after the compute, just before the
 // barrier).  The full-tile path carries no per-load predicate (a
 // runtime select around loads makes hipcc branch + drain per element
 // — guide trap 4c); only the M-tail takes the predicated path.
-template <int PASSES>
+// NC = panels-per-tile (columns/16); index decomposition: each 16-B
+// move covers half a panel row.
+template <int PASSES, int NC>
 __device__ __forceinline__ void stage_load(
     const bf16* __restrict__ src, long row0, long rows, int ld, int c0,
     int tid, bf16x8* v) {
 #pragma unroll
   for (int pass = 0; pass < PASSES; pass++) {
     int idx = tid + pass * 256;
-    int m = idx >> 3;
-    int half = idx & 7;
+    int m = idx / (NC * 2);
+    int half = idx % (NC * 2);
     int panel = half >> 1;
     int off8 = (half & 1) * 8;
     v[pass] = *(const bf16x8*)(src + (row0 + m) * (long)ld + c0 +
@@ -61,18 +62,18 @@ __device__ __forceinline__ void stage_load(
   }
 }
 
-template <int PASSES>
+template <int PASSES, int NC>
 __device__ __forceinline__ void stage_load_tail(
     const bf16* __restrict__ src, long row0, long rows, int ld, int c0,
     int tid, bf16x8* v) {
 #pragma unroll
   for (int pass = 0; pass < PASSES; pass++) {
     int idx = tid + pass * 256;
-    int m = idx >> 3;
+    int m = idx / (NC * 2);
     bf16x8 z = {};
     v[pass] = z;
     if (m < rows) {
-      int half = idx & 7;
+      int half = idx % (NC * 2);
       int panel = half >> 1;
       int off8 = (half & 1) * 8;
       v[pass] = *(const bf16x8*)(src + (row0 + m) * (long)ld + c0 +
@@ -81,14 +82,14 @@ __device__ __forceinline__ void stage_load_tail(
   }
 }
 
-template <int PASSES>
+template <int PASSES, int NC, int PANEL>
 __device__ __forceinline__ void stage_write(
     __attribute__((address_space(3))) bf16* dst, int tid, const bf16x8* v) {
 #pragma unroll
   for (int pass = 0; pass < PASSES; pass++) {
     int idx = tid + pass * 256;
-    int m = idx >> 3;
-    int half = idx & 7;
+    int m = idx / (NC * 2);
+    int half = idx % (NC * 2);
     int panel = half >> 1;
     int off8 = (half & 1) * 8;
     *(__attribute__((address_space(3))) bf16x8*)
@@ -98,6 +99,7 @@ __device__ __forceinline__ void stage_write(
 
 // fragment read: 8 m-contiguous bf16 for this lane's column, rows
 // g*8..g*8+7 of panel p — two hardware-transpose reads of [4][16]
+template <int PANEL>
 __device__ __forceinline__ bf16x8 frag_tr(
     const __attribute__((address_space(3))) bf16* tile, int panel,
     int mbase, int lane) {
@@ -116,78 +118,91 @@ __device__ __forceinline__ bf16x8 frag_tr(
   return out;
 }
 
+// NP/KP: 16-col panels per dY/X tile (tile dims = NP*16 x KP*16);
+// MT_: m rows staged per iteration.  64x64 tile: NP=KP=4, MT_=128;
+// 128x128: NP=KP=8, MT_=64 (both 64 KB LDS double-buffered).
+template <int NP, int KP, int MT_>
 __global__ __launch_bounds__(256)
 void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                   float* __restrict__ dW, long M, int N, int K,
                   long m_per_block) {
-  __shared__ bf16 smem[2 * 2 * TILE];  // [buf][dY|X][tile]
+  constexpr int PANEL = MT_ * 16;
+  constexpr int TILEY = NP * PANEL;
+  constexpr int TILEX = KP * PANEL;
+  constexpr int PY = NP * 16 * MT_ / (256 * 8);
+  constexpr int PX = KP * 16 * MT_ / (256 * 8);
+  constexpr int FN = NP / 2;           // per-wave n fragments
+  constexpr int FK = KP / 2;           // per-wave k fragments
+  __shared__ bf16 smem[2 * (TILEY + TILEX)];  // [buf][dY|X]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wn = (wid >> 1) * 32;      // wave's n offset in the tile
-  const int wk = (wid & 1) * 32;       // wave's k offset
-  const int n0 = blockIdx.y * NT;
-  const int k0 = blockIdx.x * KT;
+  const int wn = (wid >> 1) * (NP * 8);  // wave's n offset in the tile
+  const int wk = (wid & 1) * (KP * 8);   // wave's k offset
+  const int n0 = blockIdx.y * NP * 16;
+  const int k0 = blockIdx.x * KP * 16;
   long row0 = (long)blockIdx.z * m_per_block;
   long row1 = row0 + m_per_block;
   if (row1 > M) row1 = M;
   if (row0 >= row1) return;
 
   auto dyt = [&](int buf) {
-    return (__attribute__((address_space(3))) bf16*)smem + buf * 2 * TILE;
+    return (__attribute__((address_space(3))) bf16*)smem +
+           buf * (TILEY + TILEX);
   };
   auto xt = [&](int buf) {
-    return (__attribute__((address_space(3))) bf16*)smem + buf * 2 * TILE +
-           TILE;
+    return (__attribute__((address_space(3))) bf16*)smem +
+           buf * (TILEY + TILEX) + TILEY;
   };
 
-  f32x4 acc[2][2] = {};
+  f32x4 acc[FN][FK] = {};
   const int g = lane >> 4;             // 8-m chunk selector
 
-  bf16x8 vy[PASSES], vx[PASSES];
+  bf16x8 vy[PY], vx[PX];
   auto load_chunk = [&](long r) {
     long rows = row1 - r;
-    if (rows >= MT) {
-      stage_load<PASSES>(dY, r, rows, N, n0, tid, vy);
-      stage_load<PASSES>(X, r, rows, K, k0, tid, vx);
+    if (rows >= MT_) {
+      stage_load<PY, NP>(dY, r, rows, N, n0, tid, vy);
+      stage_load<PX, KP>(X, r, rows, K, k0, tid, vx);
     } else {
-      stage_load_tail<PASSES>(dY, r, rows, N, n0, tid, vy);
-      stage_load_tail<PASSES>(X, r, rows, K, k0, tid, vx);
+      stage_load_tail<PY, NP>(dY, r, rows, N, n0, tid, vy);
+      stage_load_tail<PX, KP>(X, r, rows, K, k0, tid, vx);
     }
   };
 
   int buf = 0;
   load_chunk(row0);
-  stage_write<PASSES>(dyt(buf), tid, vy);
-  stage_write<PASSES>(xt(buf), tid, vx);
-  if (row0 + MT < row1) load_chunk(row0 + MT);  // issue next loads early
+  stage_write<PY, NP, PANEL>(dyt(buf), tid, vy);
+  stage_write<PX, KP, PANEL>(xt(buf), tid, vx);
+  if (row0 + MT_ < row1) load_chunk(row0 + MT_);  // issue next loads early
   __syncthreads();
 
-  for (long r = row0; r < row1; r += MT) {
+  for (long r = row0; r < row1; r += MT_) {
     // MFMA steps over the resident tile; the NEXT chunk's global
     // loads are already in flight (issued before the barrier)
 #pragma unroll
-    for (int ms = 0; ms < MT / 32; ms++) {
+    for (int ms = 0; ms < MT_ / 32; ms++) {
       int mbase = ms * 32 + g * 8;
-      bf16x8 a[2], b[2];
+      bf16x8 a[FN], b[FK];
 #pragma unroll
-      for (int i = 0; i < 2; i++) {
-        a[i] = frag_tr(dyt(buf), (wn >> 4) + i, mbase, lane);  // dY^T
-        b[i] = frag_tr(xt(buf), (wk >> 4) + i, mbase, lane);   // X
-      }
+      for (int i = 0; i < FN; i++)
+        a[i] = frag_tr<PANEL>(dyt(buf), (wn >> 4) + i, mbase, lane);  // dY^T
 #pragma unroll
-      for (int i = 0; i < 2; i++)
+      for (int j = 0; j < FK; j++)
+        b[j] = frag_tr<PANEL>(xt(buf), (wk >> 4) + j, mbase, lane);   // X
 #pragma unroll
-        for (int j = 0; j < 2; j++)
+      for (int i = 0; i < FN; i++)
+#pragma unroll
+        for (int j = 0; j < FK; j++)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
-    long rn = r + MT;
+    long rn = r + MT_;
     if (rn < row1) {
       __syncthreads();  // everyone done reading buf^1's OLD content
-      stage_write<PASSES>(dyt(buf ^ 1), tid, vy);
-      stage_write<PASSES>(xt(buf ^ 1), tid, vx);
-      if (rn + MT < row1) load_chunk(rn + MT);  // re-issue immediately
+      stage_write<PY, NP, PANEL>(dyt(buf ^ 1), tid, vy);
+      stage_write<PX, KP, PANEL>(xt(buf ^ 1), tid, vx);
+      if (rn + MT_ < row1) load_chunk(rn + MT_);  // re-issue immediately
       __syncthreads();
       buf ^= 1;
     }
@@ -195,9 +210,9 @@ void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
 
   // C-map: col = lane&15 (k), row = (lane>>4)*4 + r (n)
 #pragma unroll
-  for (int i = 0; i < 2; i++) {
+  for (int i = 0; i < FN; i++) {
 #pragma unroll
-    for (int j = 0; j < 2; j++) {
+    for (int j = 0; j < FK; j++) {
       int col = k0 + wk + j * 16 + (lane & 15);
 #pragma unroll
       for (int r = 0; r < 4; r++) {
@@ -211,21 +226,31 @@ void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
 }  // namespace
 
 bool wgrad_supported(long M, int N, int K) {
-  return N % NT == 0 && K % KT == 0;
+  return N % 64 == 0 && K % 64 == 0;
 }
 
 void launch_conv1x1_wgrad(const rbf16* dY, const rbf16* X, float* dW,
                           long M, int N, int K, hipStream_t st) {
-  int nt = N / NT, kt = K / KT;
+  // 128x128 tile halves cross-tile re-reads when both dims allow it
+  const bool big = (N % 128 == 0) && (K % 128 == 0);
+  const int tile = big ? 128 : 64;
+  const int mt = big ? 64 : 128;
+  int nt = N / tile, kt = K / tile;
   // fill the chip: ~2 blocks per CU across the split dimension
   long want = (512 + nt * kt - 1) / (nt * kt);
-  long chunks = (M + MT - 1) / MT;
+  long chunks = (M + mt - 1) / mt;
   long split = want < 1 ? 1 : (want > chunks ? chunks : want);
-  long m_per_block = ((chunks + split - 1) / split) * MT;
+  long m_per_block = ((chunks + split - 1) / split) * mt;
   split = (M + m_per_block - 1) / m_per_block;
   dim3 grid((unsigned)kt, (unsigned)nt, (unsigned)split);
-  hipLaunchKernelGGL(wgrad_kernel, grid, dim3(256), 0, st,
-                     reinterpret_cast<const bf16*>(dY),
-                     reinterpret_cast<const bf16*>(X), dW, M, N, K,
-                     m_per_block);
+  if (big)
+    hipLaunchKernelGGL((wgrad_kernel<8, 8, 64>), grid, dim3(256), 0, st,
+                       reinterpret_cast<const bf16*>(dY),
+                       reinterpret_cast<const bf16*>(X), dW, M, N, K,
+                       m_per_block);
+  else
+    hipLaunchKernelGGL((wgrad_kernel<4, 4, 128>), grid, dim3(256), 0, st,
+                       reinterpret_cast<const bf16*>(dY),
+                       reinterpret_cast<const bf16*>(X), dW, M, N, K,
+                       m_per_block);
 }
