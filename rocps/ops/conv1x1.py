@@ -114,6 +114,12 @@ _GEMM_SHAPES = {
     (1024, 512, 1), (2048, 512, 1),
 }
 
+# (cin, cout) where the hand-written wgrad kernel (csrc/kernels/
+# wgrad.hip: dual ds_read_b64_tr_b16 transpose consumption, split-M
+# fp32 atomics) measured faster than MIOpen's wrw
+# (profiles/wgrad_ab_v3.txt); elsewhere MIOpen keeps the weight grad.
+_WGRAD_SHAPES = {(64, 64), (256, 64), (256, 128), (512, 256)}
+
 
 def set_gemm_shapes(shapes) -> None:
     _GEMM_SHAPES.clear()
@@ -162,10 +168,18 @@ class _Conv1x1Fn(torch.autograd.Function):
         wo = (wd + s - 1) // s
         # dgrad: dx_rows[m, cin] = sum_cout dy[m, cout] * W[cout, cin]
         dxr = _nt(dyr, wmat.t().contiguous())
-        if rows.is_cuda:
-            # wgrad over the huge M contraction: MIOpen's wrw kernel
-            # (the fwd ran on the stride-1 subsampled xs, so wrw of a
-            # stride-1 1x1 conv over xs is exactly dW)
+        wg = os.environ.get("ROCPS_WGRAD", "auto")
+        if (rows.is_cuda and wg != "miopen"
+                and (wg == "custom" or (cin, cout) in _WGRAD_SHAPES)
+                and rows.dtype == torch.bfloat16):
+            # hand-written tall-skinny wgrad (both operands consumed
+            # via hardware-transpose reads, split-M fp32 atomics)
+            from rocps.ops import hip_ext
+            dw = hip_ext().conv1x1_wgrad(dyr, rows).to(wmat.dtype)
+        elif rows.is_cuda:
+            # MIOpen's wrw kernel (the fwd ran on the stride-1
+            # subsampled xs, so wrw of a stride-1 1x1 conv over xs is
+            # exactly dW)
             xs = (rows.reshape(n, ho, wo, cin).permute(0, 3, 1, 2)
                   .contiguous(memory_format=torch.channels_last))
             _, dw, _ = torch.ops.aten.convolution_backward(
