@@ -125,7 +125,7 @@ template <int NP, int KP, int MT_>
 __global__ __launch_bounds__(256)
 void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
                   float* __restrict__ dW, long M, int N, int K,
-                  long m_per_block) {
+                  long m_per_block, int kt, int nt) {
   constexpr int PANEL = MT_ * 16;
   constexpr int TILEY = NP * PANEL;
   constexpr int TILEX = KP * PANEL;
@@ -139,9 +139,20 @@ void wgrad_kernel(const bf16* __restrict__ dY, const bf16* __restrict__ X,
   const int wid = tid >> 6;
   const int wn = (wid >> 1) * (NP * 8);  // wave's n offset in the tile
   const int wk = (wid & 1) * (KP * 8);   // wave's k offset
-  const int n0 = blockIdx.y * NP * 16;
-  const int k0 = blockIdx.x * KP * 16;
-  long row0 = (long)blockIdx.z * m_per_block;
+  // 1-D grid with a bijective XCD remap (T1): an XCD receives
+  // contiguous runs of (n-tile, m-split) groups, so the k-tiles that
+  // re-read the SAME dY slice land on the same L2 instead of being
+  // round-robined across all eight.
+  int bid = blockIdx.x;
+  {
+    int nwg = (int)gridDim.x;
+    int q = nwg >> 3, rm = nwg & 7, xcd = bid & 7, pos = bid >> 3;
+    bid = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + pos;
+  }
+  const int k0 = (bid % kt) * KP * 16;
+  const int n0 = ((bid / kt) % nt) * NP * 16;
+  const int bz = bid / (kt * nt);
+  long row0 = (long)bz * m_per_block;
   long row1 = row0 + m_per_block;
   if (row1 > M) row1 = M;
   if (row0 >= row1) return;
@@ -242,15 +253,15 @@ void launch_conv1x1_wgrad(const rbf16* dY, const rbf16* X, float* dW,
   long split = want < 1 ? 1 : (want > chunks ? chunks : want);
   long m_per_block = ((chunks + split - 1) / split) * mt;
   split = (M + m_per_block - 1) / m_per_block;
-  dim3 grid((unsigned)kt, (unsigned)nt, (unsigned)split);
+  dim3 grid((unsigned)(kt * nt * split));
   if (big)
     hipLaunchKernelGGL((wgrad_kernel<8, 8, 64>), grid, dim3(256), 0, st,
                        reinterpret_cast<const bf16*>(dY),
                        reinterpret_cast<const bf16*>(X), dW, M, N, K,
-                       m_per_block);
+                       m_per_block, kt, nt);
   else
     hipLaunchKernelGGL((wgrad_kernel<4, 4, 128>), grid, dim3(256), 0, st,
                        reinterpret_cast<const bf16*>(dY),
                        reinterpret_cast<const bf16*>(X), dW, M, N, K,
-                       m_per_block);
+                       m_per_block, kt, nt);
 }
