@@ -115,10 +115,11 @@ _GEMM_SHAPES = {
 }
 
 # (cin, cout) where the hand-written wgrad kernel (csrc/kernels/
-# wgrad.hip: dual ds_read_b64_tr_b16 transpose consumption, split-M
-# fp32 atomics) measured faster than MIOpen's wrw
-# (profiles/wgrad_ab_v3.txt); elsewhere MIOpen keeps the weight grad.
-_WGRAD_SHAPES = {(64, 64), (256, 64), (256, 128), (512, 256)}
+# wgrad.hip: dual ds_read_b64_tr_b16 transpose consumption, T14 split
+# staging, XCD-clustered block map, split-M fp32 atomics) measured
+# faster than MIOpen's wrw (profiles/wgrad_ab_v4.txt); elsewhere
+# MIOpen keeps the weight grad.
+_WGRAD_SHAPES = {(64, 64), (64, 256), (256, 64), (256, 128), (512, 256)}
 
 
 def set_gemm_shapes(shapes) -> None:
