@@ -101,7 +101,7 @@ def main(_):
             save_checkpoint_secs=(None if FLAGS.save_checkpoint_steps else 600),
     ) as sess:
         recovery = None
-        if FLAGS.elastic and FLAGS.issync:
+        if FLAGS.elastic and FLAGS.issync and trainer.mode == "sync":
             from rocps.train.recovery import ElasticRecovery
             recovery = ElasticRecovery(
                 trainer, FLAGS.checkpoint_dir or None,

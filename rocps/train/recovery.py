@@ -42,6 +42,11 @@ class ElasticRecovery:
     def __init__(self, trainer, checkpoint_dir: Optional[str],
                  comm_timeout: float = 60.0,
                  dead_threshold: float = 6.0, join_cap: float = 45.0):
+        if getattr(trainer, "mode", "sync") != "sync":
+            raise ValueError(
+                "ElasticRecovery covers the collective sync engine; "
+                "async/semisync worker loss is tolerated by the PS shard "
+                "itself (heartbeats + rejoin, rocps/engine/async_engine)")
         self.trainer = trainer
         self.checkpoint_dir = checkpoint_dir
         self.comm_timeout = comm_timeout

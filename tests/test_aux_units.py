@@ -379,3 +379,17 @@ def test_queue_runner_producer_exception_routes_to_coordinator():
             got.append(qr.dequeue(timeout=5))
     assert coord.should_stop()
     assert got == [0, 1, 2] or len(got) <= 3
+
+
+def test_elastic_recovery_rejects_non_sync_trainer():
+    """ElasticRecovery is a collective-engine feature; a semisync
+    trainer (mode routed to the PS engine) must be rejected loudly."""
+    import pytest
+
+    from rocps.train.recovery import ElasticRecovery
+
+    class FakeTrainer:
+        mode = "async"
+
+    with pytest.raises(ValueError):
+        ElasticRecovery(FakeTrainer(), None)
