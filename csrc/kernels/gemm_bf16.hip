@@ -55,9 +55,19 @@ void gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   const int lane = tid & 63;
   const int wid = tid >> 6;
 
-  // grid: x over N tiles, y over M tiles (consecutive blocks share A rows)
-  const int n0 = blockIdx.x * BN;
-  const int m0 = blockIdx.y * BM;
+  // flat grid with the bijective XCD remap (T1): each XCD receives a
+  // contiguous run of ids; consecutive ids share the A row-panel
+  // (same m-tile, varying n), so the panel stays in ONE L2 instead of
+  // being round-robined across all eight.
+  const int ntx = N / BN;
+  int bid = blockIdx.x;
+  {
+    int nwg = (int)gridDim.x;
+    int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, pos = bid >> 3;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int n0 = (bid % ntx) * BN;
+  const int m0 = (bid / ntx) * BM;
 
   // wave sub-tile: 2x2 waves -> each 64x64
   const int wm = (wid >> 1) * 64;
@@ -162,7 +172,7 @@ void gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
 void launch_gemm_bf16(const rbf16* A, const rbf16* W, const float* bias,
                       rbf16* D, rbf16* PRE, int M, int N, int K,
                       int epilogue, hipStream_t st) {
-  dim3 grid(N / BN, M / BM);
+  dim3 grid((unsigned)((N / BN) * (M / BM)));
   const bf16* a = reinterpret_cast<const bf16*>(A);
   const bf16* w = reinterpret_cast<const bf16*>(W);
   bf16* d = reinterpret_cast<bf16*>(D);
