@@ -55,13 +55,16 @@ void gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ W,
   const int lane = tid & 63;
   const int wid = tid >> 6;
 
-  // flat grid with the bijective XCD remap (T1): each XCD receives a
-  // contiguous run of ids; consecutive ids share the A row-panel
-  // (same m-tile, varying n), so the panel stays in ONE L2 instead of
-  // being round-robined across all eight.
+  // Flat grid; on TALL grids (few n-tiles — the Linear/conv1x1
+  // shapes) the bijective XCD remap (T1) clusters each XCD on a
+  // contiguous run of ids sharing the A row-panel, keeping the panel
+  // in ONE L2 (+18% measured on 8192x768x3072,
+  // profiles/gemm_micro_r2e.txt).  On wide/square grids the default
+  // round-robin is faster (-14% at 8192^3 with the remap), so it is
+  // gated on the n-tile count.
   const int ntx = N / BN;
   int bid = blockIdx.x;
-  {
+  if (ntx <= 8) {
     int nwg = (int)gridDim.x;
     int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, pos = bid >> 3;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
