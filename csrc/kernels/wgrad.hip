@@ -36,8 +36,6 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 // — same 64 KB LDS) halves the cross-tile re-read traffic when both
 // dims are >=128.  Waves are fixed 2x2; per-wave fragments = NP/2 x
 // KP/2 of 16x16.
-template <int MT_>
-constexpr int panel_elems() { return MT_ * 16; }
 
 // T14 split staging: LOAD (issue early, before the MFMA phase hides
 // the HBM latency) and WRITE (after the compute, just before the
