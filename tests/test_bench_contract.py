@@ -58,3 +58,40 @@ def test_bench_torchrun_two_ranks_contract():
     # whole-job aggregate: global batch = per-rank byte 16 x 2 ranks
     assert rec["config"]["global_batch"] == 32
     assert rec["value"] > 0
+
+
+def test_bench_async_mode_contract():
+    """bench.py --mode async spawns PS subprocesses and prints the same
+    JSON contract (CPU: shm push/apply/pull with the fused CPU apply).
+    One internal retry with a fresh port block — the PS/control ports
+    are derived from MASTER_PORT and can collide with a previous
+    test's lingering sockets."""
+    import socket
+
+    last = None
+    for attempt in range(2):
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        env = dict(os.environ, MASTER_PORT=str(port),
+                   MASTER_ADDR="127.0.0.1")
+        try:
+            out = subprocess.run(
+                [sys.executable, os.path.join(ROOT, "bench.py"), "--model",
+                 "mnist_mlp", "--mode", "async", "--num-ps", "2", "--steps",
+                 "3", "--warmup", "1", "--batch", "16"],
+                cwd=ROOT, env=env, capture_output=True, text=True,
+                timeout=120)
+            assert out.returncode == 0, out.stderr[-3000:]
+            lines = [l for l in out.stdout.strip().splitlines()
+                     if l.startswith("{")]
+            assert len(lines) == 1, out.stdout
+            rec = json.loads(lines[0])
+            assert rec["config"]["mode"] == "async"
+            assert rec["config"]["num_ps"] == 2
+            assert rec["value"] > 0
+            return
+        except (AssertionError, subprocess.TimeoutExpired) as e:
+            last = e
+    raise last
